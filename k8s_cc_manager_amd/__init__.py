@@ -1,0 +1,40 @@
+"""k8s-cc-manager-amd — MI355X-native Kubernetes Confidential-Computing manager.
+
+A brand-new AMD-native framework with the capabilities of NVIDIA's
+k8s-cc-manager (reference: /root/reference/main.py,
+/root/reference/gpu_operator_eviction.py): a per-node DaemonSet reconciler
+that watches the node label ``amd.com/gpu.cc.mode``, and on change
+cordons the node, evicts GPU-operator components, flips the
+Confidential-Computing (SEV-SNP / TEE-IO GPU-CC) mode on every MI355X GPU
+of the node, resets + boot-waits + attests the devices with a hand-written
+HIP/CDNA4 MFMA probe, then uncordons / reschedules and reports state back
+through node labels.
+
+Architecture (MI355X-first, not a port):
+
+- ``device/``   L1: the 14-symbol device contract (enumerate, capability
+                query, CC/fabric mode query+stage, reset, boot-wait, typed
+                errors), a fault-injectable mock, an amdsmi-backed real
+                backend and a native C++ PCI/KFD library.
+- ``k8s/``      L2: a minimal self-contained Kubernetes REST client
+                (no external k8s SDK), the component-eviction label
+                algebra, status labels, cordon/uncordon, and an
+                in-process fake API server for CPU-only testing.
+- ``core/``     L3: the reconcile manager + the 4-phase transition engine
+                (fabric-off first -> stage-all -> reset-all -> verify-all),
+                run CONCURRENTLY across the GPUs of a node (the reference
+                loops sequentially: /root/reference/main.py:486-529).
+- ``parallel/`` device executor (thread pool) + fabric barrier for
+                xGMI-hive-wide transitions.
+- ``ops/``      the one CDNA4 kernel: a post-reset MFMA+LDS attestation
+                probe (gfx950) gating ``cc.ready.state=true``.
+"""
+
+__version__ = "0.1.0"
+
+from .labels import (  # noqa: F401
+    CC_MODE_LABEL,
+    CC_STATE_LABEL,
+    CC_READY_LABEL,
+    VALID_MODES,
+)

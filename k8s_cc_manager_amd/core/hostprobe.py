@@ -1,0 +1,46 @@
+"""Host TEE capability probe.
+
+Reference behavior (/root/reference/main.py:80-103): the default mode is
+downgraded to 'off' when the host itself cannot launch confidential
+VMs. For MI355X GPU-CC the host side is AMD SEV-SNP (TEE-IO device
+interface hangs off an SNP guest context); Intel TDX hosts can also
+front TEE-IO devices, so both kvm parameter files are consulted.
+"""
+
+from __future__ import annotations
+
+import logging
+import os
+
+logger = logging.getLogger(__name__)
+
+SNP_PARAM = "/sys/module/kvm_amd/parameters/sev_snp"
+SEV_PARAM = "/sys/module/kvm_amd/parameters/sev"
+TDX_PARAM = "/sys/module/kvm_intel/parameters/tdx"
+
+_TRUTHY = ("y", "1")
+
+
+def _param_enabled(path: str) -> bool:
+    if not os.path.exists(path):
+        return False
+    try:
+        with open(path) as f:
+            return f.read().strip().lower() in _TRUTHY
+    except OSError as e:  # pragma: no cover
+        logger.warning("could not read %s: %s", path, e)
+        return False
+
+
+def is_host_cc_enabled(root: str = "") -> bool:
+    """True when the host kernel can run SNP (or TDX) guests.
+
+    ``root`` prefixes the sysfs paths for tests.
+    """
+    if _param_enabled(root + SNP_PARAM):
+        logger.info("host CC: AMD SEV-SNP enabled")
+        return True
+    if _param_enabled(root + TDX_PARAM):
+        logger.info("host CC: Intel TDX enabled")
+        return True
+    return False

@@ -1,0 +1,328 @@
+"""The reconcile core: label-driven CC-mode management for one node.
+
+Behavioral equivalent of the reference's ``CCManager``
+(/root/reference/main.py:105-695), rebuilt MI355X-first:
+
+- device work is concurrent (TransitionEngine);
+- eviction is wrapped in cordon/uncordon and unwinds on failure;
+- the watch loop's reconnect path is correct (the reference's
+  ``time.sleep(5)`` at main.py:684 NameErrors — ``time`` is never
+  imported there);
+- ERROR watch events carrying code 410 resync like HTTP 410 does;
+- transitions are timed per phase and exported as metrics.
+"""
+
+from __future__ import annotations
+
+import logging
+import threading
+import time
+from dataclasses import dataclass, field
+from typing import Optional
+
+from ..device.contract import DeviceBackend
+from ..k8s import eviction
+from ..k8s.client import ApiError, K8sClient
+from ..labels import (
+    CC_MODE_LABEL,
+    MODE_OFF,
+    MODE_PPCIE,
+    STATE_FAILED,
+    VALID_MODES,
+)
+from ..utils.metrics import METRICS
+from ..utils.readiness import create_readiness_file
+from .transition import TransitionEngine, TransitionReport
+
+logger = logging.getLogger(__name__)
+
+
+class FatalConfigError(Exception):
+    """Unrecoverable node configuration (e.g. mixed CC capability);
+    the process must exit non-zero so Kubernetes restarts/alerts
+    (reference: sys.exit(1) at main.py:240,282)."""
+
+
+@dataclass
+class ManagerConfig:
+    operator_namespace: str = "amd-gpu-operator"
+    evict_components: bool = True
+    cordon_node: bool = True
+    eviction_timeout: float = 300.0
+    eviction_poll_interval: float = 2.0
+    watch_timeout_seconds: int = 300
+    max_consecutive_errors: int = 10
+    reconnect_backoff: float = 5.0
+    readiness_file: Optional[str] = None
+
+    @classmethod
+    def from_env(cls, env=None) -> "ManagerConfig":
+        import os
+
+        env = env if env is not None else os.environ
+        return cls(
+            operator_namespace=env.get("OPERATOR_NAMESPACE", "amd-gpu-operator"),
+            evict_components=env.get("EVICT_OPERATOR_COMPONENTS", "true").lower()
+            == "true",
+            cordon_node=env.get("CORDON_NODE", "true").lower() == "true",
+            readiness_file=env.get("CC_READINESS_FILE"),
+        )
+
+
+class CCManager:
+    def __init__(
+        self,
+        node_name: str,
+        default_mode: str,
+        host_cc: bool,
+        k8s: K8sClient,
+        backend: DeviceBackend,
+        engine: Optional[TransitionEngine] = None,
+        config: Optional[ManagerConfig] = None,
+    ):
+        self.node_name = node_name
+        self.default_mode = default_mode
+        self.host_cc = host_cc
+        self.k8s = k8s
+        self.backend = backend
+        self.engine = engine or TransitionEngine()
+        self.config = config or ManagerConfig()
+        self.current_label: Optional[str] = None
+        self.current_rv: Optional[str] = None
+        self.last_report: Optional[TransitionReport] = None
+        self._transition_lock = threading.Lock()
+        self.stop_event = threading.Event()
+
+    # ------------------------------------------------------------------
+    # label plumbing
+    # ------------------------------------------------------------------
+    def with_default(self, label: Optional[str]) -> str:
+        if not label:
+            logger.info("no %s label; applying default %r", CC_MODE_LABEL, self.default_mode)
+            return self.default_mode
+        return label
+
+    def read_mode_label(self) -> str:
+        """Read the desired-mode label + resourceVersion from the node."""
+        node = self.k8s.get_node(self.node_name)
+        labels = (node.get("metadata") or {}).get("labels") or {}
+        self.current_label = labels.get(CC_MODE_LABEL, "")
+        self.current_rv = (node.get("metadata") or {}).get("resourceVersion")
+        return self.current_label
+
+    def _set_state(self, state: str) -> None:
+        eviction.set_cc_state_label(self.k8s, self.node_name, state)
+
+    # ------------------------------------------------------------------
+    # mode application (reference dispatcher: main.py:214-263)
+    # ------------------------------------------------------------------
+    def apply_mode(self, mode: str) -> bool:
+        with self._transition_lock:
+            return self._apply_mode_locked(mode)
+
+    def _apply_mode_locked(self, mode: str) -> bool:
+        if mode and mode not in VALID_MODES:
+            logger.error("invalid CC mode %r (valid: %s)", mode, ",".join(VALID_MODES))
+            self._set_state(STATE_FAILED)
+            return False
+        if not self.host_cc and mode not in ("", MODE_OFF):
+            logger.warning("host is not CC-capable but GPU mode %r requested", mode)
+
+        if mode == MODE_PPCIE:
+            return self._apply_fabric()
+
+        gpus = self.backend.get_gpus()
+        cc_gpus = self.backend.get_cc_capable_gpus()
+
+        # Mixed capability with a non-off target is unrecoverable
+        # (reference main.py:237-240).
+        if mode and mode != MODE_OFF and len(gpus) != len(cc_gpus):
+            missing = {g.bdf for g in gpus} - {g.bdf for g in cc_gpus}
+            raise FatalConfigError(f"GPUs without CC support: {sorted(missing)}")
+
+        if not gpus:
+            logger.warning("no GPUs to configure")
+            return True
+        if not mode:
+            logger.info("no CC mode specified, skipping")
+            return True
+        if not cc_gpus:
+            self._set_state(MODE_OFF)
+            return True
+
+        if self._cc_mode_is_set(cc_gpus, mode):
+            logger.info("all GPUs already in CC mode %r", mode)
+            self._set_state(mode)
+            return True
+
+        devices, _ = self.backend.find_devices()
+        runner = lambda: self.engine.apply_cc_mode(devices, cc_gpus, mode)  # noqa: E731
+        if self.config.evict_components:
+            return self._run_with_eviction(mode, runner)
+        return self._run_direct(mode, runner)
+
+    def _apply_fabric(self) -> bool:
+        devices, _ = self.backend.find_devices()
+        capable = self.backend.get_fabric_capable_devices()
+        if len(devices) != len(capable):
+            missing = {d.bdf for d in devices} - {d.bdf for d in capable}
+            raise FatalConfigError(f"devices without fabric-mode support: {sorted(missing)}")
+        if not devices:
+            logger.warning("no devices to configure for fabric mode")
+            return True
+        if self._fabric_mode_is_set(devices):
+            logger.info("all devices already in fabric-protected mode")
+            self._set_state(MODE_PPCIE)
+            return True
+        runner = lambda: self.engine.apply_fabric_mode(devices)  # noqa: E731
+        if self.config.evict_components:
+            return self._run_with_eviction(MODE_PPCIE, runner)
+        return self._run_direct(MODE_PPCIE, runner)
+
+    def _cc_mode_is_set(self, gpus, mode: str) -> bool:
+        try:
+            return all(g.query_cc_mode() == mode for g in gpus)
+        except Exception as e:
+            logger.error("CC mode pre-check failed: %s", e)
+            return False
+
+    def _fabric_mode_is_set(self, devices) -> bool:
+        try:
+            return all(d.query_fabric_mode() == "on" for d in devices)
+        except Exception as e:
+            logger.error("fabric mode pre-check failed: %s", e)
+            return False
+
+    # ------------------------------------------------------------------
+    # transition wrappers
+    # ------------------------------------------------------------------
+    def _run_direct(self, mode: str, runner) -> bool:
+        report: TransitionReport = runner()
+        self.last_report = report
+        METRICS.observe_transition(mode, report.ok, report.seconds, report.phases)
+        self._set_state(mode if report.ok else STATE_FAILED)
+        return report.ok
+
+    def _run_with_eviction(self, mode: str, runner) -> bool:
+        """cordon -> snapshot labels -> evict -> transition -> reschedule
+        -> uncordon. Unwinds on eviction failure (reference gap:
+        main.py:558-566)."""
+        cfg = self.config
+        cordoned = False
+        if cfg.cordon_node:
+            cordoned = eviction.cordon(self.k8s, self.node_name)
+
+        try:
+            snapshot = eviction.fetch_component_labels(self.k8s, self.node_name)
+        except ApiError as e:
+            logger.error("could not snapshot component labels: %s", e)
+            if cordoned:
+                eviction.uncordon(self.k8s, self.node_name)
+            return False
+
+        if not eviction.evict_components(
+            self.k8s,
+            self.node_name,
+            cfg.operator_namespace,
+            snapshot,
+            timeout=cfg.eviction_timeout,
+            poll_interval=cfg.eviction_poll_interval,
+        ):
+            logger.error("eviction failed; unwinding")
+            eviction.unwind_paused_labels(self.k8s, self.node_name, snapshot)
+            if cordoned:
+                eviction.uncordon(self.k8s, self.node_name)
+            return False
+
+        ok = self._run_direct(mode, runner)
+
+        if not eviction.reschedule_components(self.k8s, self.node_name, snapshot):
+            logger.error("failed to reschedule operator components")
+            ok = False
+        if cordoned:
+            eviction.uncordon(self.k8s, self.node_name)
+        return ok
+
+    # ------------------------------------------------------------------
+    # watch loop (reference: main.py:600-684, with the bugs fixed)
+    # ------------------------------------------------------------------
+    def run(self) -> None:
+        self.read_mode_label()
+        self.apply_mode(self.with_default(self.current_label))
+        create_readiness_file(self.config.readiness_file)
+
+        last_applied = self.current_label
+        consecutive_errors = 0
+        logger.info(
+            "watching node %r for %s (current=%r, rv=%s)",
+            self.node_name,
+            CC_MODE_LABEL,
+            self.current_label,
+            self.current_rv,
+        )
+        while not self.stop_event.is_set():
+            try:
+                resync = False
+                for event in self.k8s.watch_node(
+                    self.node_name,
+                    resource_version=self.current_rv,
+                    timeout_seconds=self.config.watch_timeout_seconds,
+                ):
+                    if self.stop_event.is_set():
+                        return
+                    etype = event.get("type")
+                    obj = event.get("object") or {}
+                    if etype == "ERROR":
+                        code = obj.get("code")
+                        logger.error("watch ERROR event: %s", obj)
+                        if code == 410:
+                            resync = True
+                        else:
+                            consecutive_errors += 1
+                        break
+                    consecutive_errors = 0
+                    meta = obj.get("metadata") or {}
+                    if meta.get("resourceVersion"):
+                        self.current_rv = meta["resourceVersion"]
+                    if etype == "BOOKMARK":
+                        continue
+                    if etype in ("ADDED", "MODIFIED"):
+                        labels = meta.get("labels") or {}
+                        self.current_label = labels.get(CC_MODE_LABEL, "")
+                        if self.current_label != last_applied:
+                            logger.info(
+                                "label changed %r -> %r", last_applied, self.current_label
+                            )
+                            last_applied = self.current_label
+                            self.apply_mode(self.with_default(self.current_label))
+                if resync:
+                    last_applied = self._resync(last_applied)
+            except ApiError as e:
+                consecutive_errors += 1
+                if consecutive_errors >= self.config.max_consecutive_errors:
+                    raise RuntimeError(
+                        f"watch failed {consecutive_errors} times consecutively: {e}"
+                    )
+                if e.status == 410:
+                    logger.warning("resourceVersion %s expired (410); resync", self.current_rv)
+                    last_applied = self._resync(last_applied)
+                logger.info("reconnecting watch in %.0fs", self.config.reconnect_backoff)
+                self._sleep(self.config.reconnect_backoff)
+            except Exception as e:
+                consecutive_errors += 1
+                if consecutive_errors >= self.config.max_consecutive_errors:
+                    raise
+                logger.error("watch stream error: %s; reconnecting", e)
+                self._sleep(self.config.reconnect_backoff)
+
+    def _resync(self, last_applied: Optional[str]) -> Optional[str]:
+        """Full re-list after RV compaction (reference main.py:670-682)."""
+        self.read_mode_label()
+        if self.current_label != last_applied:
+            logger.info("resync: label %r -> %r", last_applied, self.current_label)
+            last_applied = self.current_label
+            self.apply_mode(self.with_default(self.current_label))
+        return last_applied
+
+    def _sleep(self, seconds: float) -> None:
+        self.stop_event.wait(timeout=seconds)

@@ -1,0 +1,241 @@
+"""The 4-phase CC / fabric mode transition engine.
+
+Semantics follow the reference's two transition machines
+(/root/reference/main.py:449-542 CC, :317-391 PPCIe/fabric) — the
+ordering invariants are load-bearing and kept exactly:
+
+1. fabric-protected mode is forced OFF everywhere before any CC change;
+2. the new mode is STAGED on every device before any reset;
+3. all staged devices reset together;
+4. every device boot-waits, its mode readback is verified, and (new
+   here) a HIP/CDNA4 attestation probe must pass before the node can be
+   labeled ready.
+
+What is different (MI355X-first):
+
+- every phase fans out CONCURRENTLY over the node's GPUs
+  (:class:`~k8s_cc_manager_amd.parallel.DeviceExecutor`); the reference
+  loops serially;
+- an optional cross-process fabric barrier guards the stage->reset seam
+  when ranks own individual GPUs (bench scaling harness);
+- per-phase wall-clock is recorded (the BASELINE metric).
+"""
+
+from __future__ import annotations
+
+import logging
+from dataclasses import dataclass, field
+from typing import Callable, Dict, List, Optional, Sequence
+
+from ..device.contract import (
+    CCDevice,
+    CCDeviceError,
+    FABRIC_OFF,
+    FABRIC_ON,
+    ModeVerifyError,
+)
+from ..parallel.executor import DeviceExecutor, PerDeviceError
+from ..utils.timing import PhaseTimer
+
+logger = logging.getLogger(__name__)
+
+#: signature of the post-reset attestation hook: device -> None (raises on failure)
+Attestor = Callable[[CCDevice], None]
+
+
+class TransitionError(Exception):
+    """A transition failed; the node must be labeled 'failed'."""
+
+
+@dataclass
+class TransitionReport:
+    ok: bool
+    mode: str
+    seconds: float = 0.0
+    phases: Dict[str, float] = field(default_factory=dict)
+    devices_changed: List[str] = field(default_factory=list)
+    error: str = ""
+
+
+class TransitionEngine:
+    def __init__(
+        self,
+        executor: Optional[DeviceExecutor] = None,
+        attestor: Optional[Attestor] = None,
+        boot_timeout: float = 60.0,
+        barrier: Optional[object] = None,  # needs .wait(timeout=...)
+    ):
+        self.executor = executor or DeviceExecutor()
+        self.attestor = attestor
+        self.boot_timeout = boot_timeout
+        self.barrier = barrier
+
+    # ------------------------------------------------------------------
+    def _barrier_wait(self) -> None:
+        if self.barrier is not None:
+            self.barrier.wait()
+
+    def _boot_verify(
+        self, dev: CCDevice, query: Callable[[CCDevice], str], want: str, what: str
+    ) -> None:
+        dev.wait_for_boot(timeout=self.boot_timeout)
+        got = query(dev)
+        if got != want:
+            raise ModeVerifyError(
+                f"{dev.bdf}: {what} mode readback {got!r} != staged {want!r}"
+            )
+        if self.attestor is not None:
+            self.attestor(dev)
+
+    # ------------------------------------------------------------------
+    def apply_cc_mode(
+        self,
+        all_devices: Sequence[CCDevice],
+        gpus: Sequence[CCDevice],
+        mode: str,
+    ) -> TransitionReport:
+        """Set CC mode on ``gpus``; ``all_devices`` is consulted for the
+        fabric-off precondition (reference main.py:471-500)."""
+        timer = PhaseTimer()
+        report = TransitionReport(ok=False, mode=mode)
+        logger.info("applying CC mode %r on %d GPU(s)", mode, len(gpus))
+        try:
+            # Phase 1: fabric-protected mode off everywhere ------------
+            timer.start("fabric-off")
+            fabric_devs = [d for d in all_devices if d.fabric_query_supported]
+            fabric_on, _ = self.executor.partition(
+                "fabric-query", fabric_devs, lambda d: d.query_fabric_mode() != FABRIC_OFF
+            )
+            if fabric_on:
+                logger.info("disabling fabric mode on %d device(s)", len(fabric_on))
+                self.executor.run(
+                    "fabric-stage-off", fabric_on, lambda d: d.set_fabric_mode(FABRIC_OFF)
+                )
+                self._barrier_wait()
+                self.executor.run("fabric-reset", fabric_on, lambda d: d.reset())
+                self.executor.run(
+                    "fabric-verify",
+                    fabric_on,
+                    lambda d: self._boot_verify(
+                        d, lambda x: x.query_fabric_mode(), FABRIC_OFF, "fabric"
+                    ),
+                )
+
+            # Phase 2: stage CC mode on GPUs not already there ---------
+            timer.start("stage")
+            to_change, already = self.executor.partition(
+                "cc-query", gpus, lambda d: d.query_cc_mode() != mode
+            )
+            for d in already:
+                logger.info("%s already in CC mode %r", d.bdf, mode)
+            self.executor.run("cc-stage", to_change, lambda d: d.set_cc_mode(mode))
+            report.devices_changed = [d.bdf for d in to_change]
+
+            # Phase 3: reset all staged GPUs together ------------------
+            self._barrier_wait()
+            timer.start("reset")
+            self.executor.run("cc-reset", to_change, lambda d: d.reset())
+
+            # Phase 4: boot-wait + verify + attest ---------------------
+            timer.start("verify")
+            self.executor.run(
+                "cc-verify",
+                to_change,
+                lambda d: self._boot_verify(
+                    d, lambda x: x.query_cc_mode(), mode, "cc"
+                ),
+            )
+            timer.stop()
+        except (PerDeviceError, CCDeviceError, Exception) as e:
+            timer.stop()
+            report.seconds = timer.total()
+            report.phases = timer.as_dict()
+            report.error = str(e)
+            logger.error("CC transition to %r failed: %s", mode, e)
+            return report
+        report.ok = True
+        report.seconds = timer.total()
+        report.phases = timer.as_dict()
+        logger.info(
+            "CC mode %r applied to %d GPU(s) in %.3fs (%s)",
+            mode,
+            len(gpus),
+            report.seconds,
+            ", ".join(f"{k}={v:.3f}s" for k, v in report.phases.items()),
+        )
+        return report
+
+    # ------------------------------------------------------------------
+    def apply_fabric_mode(self, devices: Sequence[CCDevice]) -> TransitionReport:
+        """Enable the fabric-protected (xGMI-hive) mode on all devices.
+
+        Mirrors the reference's PPCIe machine (main.py:317-391): force
+        off first (per-device reset), stage ON everywhere, reset the
+        hive together, verify all.
+        """
+        timer = PhaseTimer()
+        report = TransitionReport(ok=False, mode="ppcie")
+        logger.info("applying fabric-protected mode on %d device(s)", len(devices))
+        try:
+            # Phase 1: force fabric mode off where it is on ------------
+            timer.start("force-off")
+            stuck_on, _ = self.executor.partition(
+                "fabric-query", devices, lambda d: d.query_fabric_mode() != FABRIC_OFF
+            )
+            if stuck_on:
+                self.executor.run(
+                    "fabric-off-stage", stuck_on, lambda d: d.set_fabric_mode(FABRIC_OFF)
+                )
+                self._barrier_wait()
+                self.executor.run("fabric-off-reset", stuck_on, lambda d: d.reset())
+                self.executor.run(
+                    "fabric-off-verify",
+                    stuck_on,
+                    lambda d: self._boot_verify(
+                        d, lambda x: x.query_fabric_mode(), FABRIC_OFF, "fabric"
+                    ),
+                )
+
+            # Phase 2: stage ON on ALL devices (no reset yet) ----------
+            timer.start("stage")
+            to_change, already = self.executor.partition(
+                "fabric-query2", devices, lambda d: d.query_fabric_mode() != FABRIC_ON
+            )
+            for d in already:
+                logger.info("%s already in fabric-protected mode", d.bdf)
+            self.executor.run(
+                "fabric-stage", to_change, lambda d: d.set_fabric_mode(FABRIC_ON)
+            )
+            report.devices_changed = [d.bdf for d in to_change]
+
+            # Phase 3: reset the hive together -------------------------
+            self._barrier_wait()
+            timer.start("reset")
+            self.executor.run("fabric-reset", to_change, lambda d: d.reset())
+
+            # Phase 4: verify + attest ---------------------------------
+            timer.start("verify")
+            self.executor.run(
+                "fabric-verify",
+                to_change,
+                lambda d: self._boot_verify(
+                    d, lambda x: x.query_fabric_mode(), FABRIC_ON, "fabric"
+                ),
+            )
+            timer.stop()
+        except (PerDeviceError, CCDeviceError, Exception) as e:
+            timer.stop()
+            report.seconds = timer.total()
+            report.phases = timer.as_dict()
+            report.error = str(e)
+            logger.error("fabric transition failed: %s", e)
+            return report
+        report.ok = True
+        report.seconds = timer.total()
+        report.phases = timer.as_dict()
+        logger.info(
+            "fabric-protected mode applied in %.3fs (%s)",
+            report.seconds,
+            ", ".join(f"{k}={v:.3f}s" for k, v in report.phases.items()),
+        )
+        return report

@@ -1,0 +1,148 @@
+"""The device-layer contract.
+
+This is the exact capability surface the reference consumes from
+gpu-admin-tools (14 symbols, catalogued in SURVEY.md §1/L1 from
+/root/reference/main.py:144-212,298-542), re-expressed AMD-native:
+
+==========================  ===========================================
+reference (NVIDIA)          this contract (MI355X)
+==========================  ===========================================
+``find_gpus()``             :meth:`DeviceBackend.find_devices`
+``.bdf`` / ``.name``        :attr:`CCDevice.bdf` / :attr:`CCDevice.name`
+``.is_gpu()``               :meth:`CCDevice.is_gpu`
+``.is_nvswitch()``          :meth:`CCDevice.is_fabric_switch` (always
+                            False on MI355X — xGMI is point-to-point,
+                            there is no switch device; kept so the
+                            reconcile core stays shape-compatible)
+``.is_cc_query_supported``  :attr:`CCDevice.cc_query_supported`
+``.is_ppcie_query_...``     :attr:`CCDevice.fabric_query_supported`
+``.query_cc_mode()``        :meth:`CCDevice.query_cc_mode`
+``.set_cc_mode(m)``         :meth:`CCDevice.set_cc_mode` (STAGED —
+                            takes effect at the next reset, like the
+                            reference's register write, main.py:502-512)
+``.query_ppcie_mode()``     :meth:`CCDevice.query_fabric_mode`
+``.set_ppcie_mode(m)``      :meth:`CCDevice.set_fabric_mode` (staged)
+``.reset_with_os()``        :meth:`CCDevice.reset`
+``.wait_for_boot()``        :meth:`CCDevice.wait_for_boot`
+``GpuError``                :exc:`CCDeviceError`
+==========================  ===========================================
+
+All methods may be called from multiple threads: the reconcile core runs
+per-device transitions concurrently (one worker per GPU of the node);
+implementations must be thread-safe per device.
+"""
+
+from __future__ import annotations
+
+import abc
+from typing import List, Tuple
+
+# Fabric-protected mode values (the reference's PPCIe vocabulary,
+# main.py:341-359).
+FABRIC_OFF = "off"
+FABRIC_ON = "on"
+
+
+class CCDeviceError(Exception):
+    """Base error of the device layer (analogue of gpu-admin-tools'
+    ``GpuError``, /root/reference/main.py:40)."""
+
+
+class ResetError(CCDeviceError):
+    """Device reset failed or the device fell off the bus."""
+
+
+class BootTimeoutError(CCDeviceError):
+    """Device did not come back within the boot-wait deadline."""
+
+
+class ModeVerifyError(CCDeviceError):
+    """Post-reset mode readback disagreed with the staged mode."""
+
+
+class CCDevice(abc.ABC):
+    """One managed device (an MI355X GPU)."""
+
+    #: PCI bus:device.function, e.g. "0000:0a:00.0"
+    bdf: str
+    #: Human-readable device name, e.g. "AMD Instinct MI355X"
+    name: str
+
+    # -- classification -------------------------------------------------
+    @abc.abstractmethod
+    def is_gpu(self) -> bool:
+        ...
+
+    def is_fabric_switch(self) -> bool:
+        """MI355X nodes have no fabric-switch PCI device (xGMI is p2p,
+        7 links x ~153 GB/s per GPU); the hook exists for contract parity
+        with the reference's NVSwitch handling (main.py:167-175)."""
+        return False
+
+    # -- capability -----------------------------------------------------
+    @property
+    @abc.abstractmethod
+    def cc_query_supported(self) -> bool:
+        ...
+
+    @property
+    @abc.abstractmethod
+    def fabric_query_supported(self) -> bool:
+        ...
+
+    # -- CC mode --------------------------------------------------------
+    @abc.abstractmethod
+    def query_cc_mode(self) -> str:
+        """Current CC mode: 'on' | 'off' | 'devtools'."""
+
+    @abc.abstractmethod
+    def set_cc_mode(self, mode: str) -> None:
+        """Stage a CC mode; applied at the next :meth:`reset`."""
+
+    # -- fabric (xGMI-hive) mode ---------------------------------------
+    @abc.abstractmethod
+    def query_fabric_mode(self) -> str:
+        """Current fabric-protected mode: 'on' | 'off'."""
+
+    @abc.abstractmethod
+    def set_fabric_mode(self, mode: str) -> None:
+        """Stage the fabric-protected mode; applied at the next reset."""
+
+    # -- lifecycle ------------------------------------------------------
+    @abc.abstractmethod
+    def reset(self) -> None:
+        """Function-level / mode-1 reset applying staged modes."""
+
+    @abc.abstractmethod
+    def wait_for_boot(self, timeout: float = 60.0) -> None:
+        """Block until the device is responsive post-reset, else raise
+        :exc:`BootTimeoutError`."""
+
+    def __repr__(self) -> str:  # pragma: no cover - cosmetic
+        return f"<{type(self).__name__} {self.bdf} {self.name!r}>"
+
+
+class DeviceBackend(abc.ABC):
+    """Factory/enumerator for :class:`CCDevice` objects."""
+
+    @abc.abstractmethod
+    def find_devices(self) -> Tuple[List[CCDevice], int]:
+        """All managed AMD devices of the node, plus their count
+        (same return shape as the reference's ``find_gpus()``,
+        main.py:144-155)."""
+
+    # Convenience partitions (reference: main.py:157-212) --------------
+    def get_gpus(self) -> List[CCDevice]:
+        devices, _ = self.find_devices()
+        return [d for d in devices if d.is_gpu()]
+
+    def get_fabric_switches(self) -> List[CCDevice]:
+        devices, _ = self.find_devices()
+        return [d for d in devices if d.is_fabric_switch()]
+
+    def get_cc_capable_gpus(self) -> List[CCDevice]:
+        return [g for g in self.get_gpus() if g.cc_query_supported]
+
+    def get_fabric_capable_devices(self) -> List[CCDevice]:
+        devices, _ = self.find_devices()
+        return [d for d in devices if d.fabric_query_supported]

@@ -1,0 +1,225 @@
+"""Minimal self-contained Kubernetes REST client.
+
+The reference depends on the official ``kubernetes`` Python SDK
+(/root/reference/requirements.txt:2) for exactly five call shapes:
+read_node, patch_node, list_namespaced_pod, list_node + watch stream
+(SURVEY.md §1/L2). This client implements those five over plain HTTPS
+with ``requests`` — no SDK, ~no transitive dependency surface (the
+reference ships 21 third-party distributions for this), and JSON dicts
+instead of generated model classes.
+
+Improvements over the reference's usage:
+
+- node label updates go through ``application/strategic-merge-patch+json``
+  on the labels map only, instead of read-modify-write of the whole Node
+  object (reference: gpu_operator_eviction.py:165-170 PATCHes the full
+  node it just read — a lost-update window);
+- cordon/uncordon (spec.unschedulable) is first-class;
+- the watch stream is a plain chunked-JSON iterator with explicit
+  bookmark support.
+"""
+
+from __future__ import annotations
+
+import json
+import logging
+import os
+from typing import Any, Dict, Iterator, Optional
+
+import requests
+
+logger = logging.getLogger(__name__)
+
+SA_DIR = "/var/run/secrets/kubernetes.io/serviceaccount"
+
+
+class ApiError(Exception):
+    """HTTP-level API failure (analogue of kubernetes.client.rest.ApiException)."""
+
+    def __init__(self, status: int, reason: str = "", body: str = ""):
+        super().__init__(f"kubernetes API error {status}: {reason}")
+        self.status = status
+        self.reason = reason
+        self.body = body
+
+
+def _raise_for(resp: requests.Response) -> None:
+    if resp.status_code >= 400:
+        raise ApiError(resp.status_code, resp.reason or "", resp.text[:2048])
+
+
+class K8sClient:
+    """Thin typed wrapper over the core/v1 REST surface we need."""
+
+    def __init__(
+        self,
+        base_url: str,
+        token: Optional[str] = None,
+        verify: Any = True,
+        cert: Any = None,
+        session: Optional[requests.Session] = None,
+    ):
+        self.base_url = base_url.rstrip("/")
+        self._session = session or requests.Session()
+        self._session.verify = verify
+        if cert:
+            self._session.cert = cert
+        if token:
+            self._session.headers["Authorization"] = f"Bearer {token}"
+        self._session.headers["Accept"] = "application/json"
+
+    # -- construction ---------------------------------------------------
+    @classmethod
+    def in_cluster(cls) -> "K8sClient":
+        host = os.environ.get("KUBERNETES_SERVICE_HOST")
+        port = os.environ.get("KUBERNETES_SERVICE_PORT", "443")
+        if not host:
+            raise RuntimeError("not running in-cluster (no KUBERNETES_SERVICE_HOST)")
+        token_path = os.path.join(SA_DIR, "token")
+        ca_path = os.path.join(SA_DIR, "ca.crt")
+        with open(token_path) as f:
+            token = f.read().strip()
+        verify: Any = ca_path if os.path.exists(ca_path) else True
+        return cls(f"https://{host}:{port}", token=token, verify=verify)
+
+    @classmethod
+    def from_kubeconfig(cls, path: Optional[str] = None) -> "K8sClient":
+        import base64
+        import tempfile
+
+        import yaml
+
+        path = path or os.environ.get("KUBECONFIG") or os.path.expanduser("~/.kube/config")
+        with open(path) as f:
+            cfg = yaml.safe_load(f)
+        ctx_name = cfg.get("current-context")
+        ctx = next(c["context"] for c in cfg["contexts"] if c["name"] == ctx_name)
+        cluster = next(c["cluster"] for c in cfg["clusters"] if c["name"] == ctx["cluster"])
+        user = next(u["user"] for u in cfg["users"] if u["name"] == ctx["user"])
+
+        def materialize(data_key: str, path_key: str) -> Optional[str]:
+            if path_key in user:
+                return user[path_key]
+            if data_key in user:
+                tf = tempfile.NamedTemporaryFile(delete=False, suffix=".pem")
+                tf.write(base64.b64decode(user[data_key]))
+                tf.close()
+                return tf.name
+            return None
+
+        verify: Any = True
+        if "certificate-authority" in cluster:
+            verify = cluster["certificate-authority"]
+        elif "certificate-authority-data" in cluster:
+            tf = tempfile.NamedTemporaryFile(delete=False, suffix=".pem")
+            tf.write(base64.b64decode(cluster["certificate-authority-data"]))
+            tf.close()
+            verify = tf.name
+        elif cluster.get("insecure-skip-tls-verify"):
+            verify = False
+
+        cert = None
+        cc = materialize("client-certificate-data", "client-certificate")
+        ck = materialize("client-key-data", "client-key")
+        if cc and ck:
+            cert = (cc, ck)
+        return cls(cluster["server"], token=user.get("token"), verify=verify, cert=cert)
+
+    # -- nodes ----------------------------------------------------------
+    def get_node(self, name: str) -> Dict[str, Any]:
+        resp = self._session.get(f"{self.base_url}/api/v1/nodes/{name}")
+        _raise_for(resp)
+        return resp.json()
+
+    def patch_node_labels(self, name: str, labels: Dict[str, Optional[str]]) -> Dict[str, Any]:
+        """Merge-patch only the given labels (None deletes a label)."""
+        patch = {"metadata": {"labels": labels}}
+        resp = self._session.patch(
+            f"{self.base_url}/api/v1/nodes/{name}",
+            data=json.dumps(patch),
+            headers={"Content-Type": "application/strategic-merge-patch+json"},
+        )
+        _raise_for(resp)
+        return resp.json()
+
+    def set_node_unschedulable(self, name: str, unschedulable: bool) -> Dict[str, Any]:
+        """Cordon (True) / uncordon (False) the node."""
+        patch = {"spec": {"unschedulable": unschedulable or None}}
+        resp = self._session.patch(
+            f"{self.base_url}/api/v1/nodes/{name}",
+            data=json.dumps(patch),
+            headers={"Content-Type": "application/strategic-merge-patch+json"},
+        )
+        _raise_for(resp)
+        return resp.json()
+
+    # -- pods -----------------------------------------------------------
+    def list_pods(
+        self,
+        namespace: str,
+        field_selector: str = "",
+        label_selector: str = "",
+    ) -> Dict[str, Any]:
+        params: Dict[str, str] = {}
+        if field_selector:
+            params["fieldSelector"] = field_selector
+        if label_selector:
+            params["labelSelector"] = label_selector
+        resp = self._session.get(
+            f"{self.base_url}/api/v1/namespaces/{namespace}/pods", params=params
+        )
+        _raise_for(resp)
+        return resp.json()
+
+    # -- watch ----------------------------------------------------------
+    def watch_node(
+        self,
+        name: str,
+        resource_version: Optional[str] = None,
+        timeout_seconds: int = 300,
+    ) -> Iterator[Dict[str, Any]]:
+        """Stream watch events for one node.
+
+        Yields dicts {"type": "ADDED|MODIFIED|DELETED|BOOKMARK|ERROR",
+        "object": {...}}. Returns when the server closes the stream
+        (after ``timeout_seconds``). HTTP-level failures raise ApiError
+        — including the 410 Gone the reconcile loop resyncs on.
+        """
+        params: Dict[str, str] = {
+            "watch": "true",
+            "fieldSelector": f"metadata.name={name}",
+            "timeoutSeconds": str(timeout_seconds),
+            "allowWatchBookmarks": "true",
+        }
+        if resource_version:
+            params["resourceVersion"] = resource_version
+        resp = self._session.get(
+            f"{self.base_url}/api/v1/nodes",
+            params=params,
+            stream=True,
+            timeout=timeout_seconds + 30,
+        )
+        _raise_for(resp)
+        try:
+            for line in resp.iter_lines():
+                if not line:
+                    continue
+                try:
+                    yield json.loads(line)
+                except json.JSONDecodeError:  # pragma: no cover - server junk
+                    logger.warning("undecodable watch line: %.200r", line)
+        finally:
+            resp.close()
+
+
+def load_client(kubeconfig: str = "") -> K8sClient:
+    """In-cluster first, kubeconfig fallback (reference order,
+    /root/reference/main.py:128-140)."""
+    try:
+        c = K8sClient.in_cluster()
+        logger.info("loaded in-cluster kubernetes configuration")
+        return c
+    except Exception:
+        c = K8sClient.from_kubeconfig(kubeconfig or None)
+        logger.info("loaded kubeconfig configuration")
+        return c

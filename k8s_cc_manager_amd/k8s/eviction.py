@@ -1,0 +1,217 @@
+"""GPU-operator component eviction + status labels + cordon.
+
+Behavioral port (not a code port) of the reference's eviction module
+(/root/reference/gpu_operator_eviction.py): before a CC-mode flip the
+node's GPU-operator components must be stopped, because a mode-1 reset
+kills every KFD process holding the device — on AMD this ordering is
+even more load-bearing than on NVIDIA (amdgpu reset vs open KFD handles,
+SURVEY.md §7 hard-parts (a)).
+
+Mechanism (same as reference): the operator deploys components according
+to per-component *node labels*; rewriting a label to its "paused" form
+makes the operator delete that component's pods on this node; restoring
+it reschedules them. The paused/unpaused value algebra preserves user
+intent ('false' and '' stay untouched; custom values get a reversible
+suffix) — semantics identical to gpu_operator_eviction.py:43-95, written
+fresh.
+
+Differences from the reference (all deliberate):
+
+- one strategic-merge patch for all labels instead of read-modify-write
+  of the whole Node object;
+- the pod-drain poll watches ALL components in one loop under a single
+  deadline (the reference serializes: up to 300 s per component);
+- eviction failures unwind the paused labels (the reference returns
+  early leaving components paused, main.py:558-566 — flagged in
+  SURVEY.md §5);
+- cordon/uncordon around the transition (north-star addition).
+"""
+
+from __future__ import annotations
+
+import logging
+import time
+from typing import Dict, Optional
+
+from ..labels import CC_READY_LABEL, CC_STATE_LABEL, ready_value_for_state
+from .client import ApiError, K8sClient
+
+logger = logging.getLogger(__name__)
+
+PAUSED_VALUE = "paused-for-cc-mode-change"
+
+# AMD GPU-operator components that must not hold the device across a
+# CC transition (reference set: gpu_operator_eviction.py:23-38).
+COMPONENT_LABELS = [
+    "amd.com/gpu.deploy.device-plugin",
+    "amd.com/gpu.deploy.node-labeller",
+    "amd.com/gpu.deploy.metrics-exporter",
+    "amd.com/gpu.deploy.vfio-manager",
+    "amd.com/gpu.deploy.test-runner",
+]
+
+COMPONENT_APP_LABELS = {
+    "amd.com/gpu.deploy.device-plugin": "amd-gpu-device-plugin",
+    "amd.com/gpu.deploy.node-labeller": "amd-gpu-node-labeller",
+    "amd.com/gpu.deploy.metrics-exporter": "amd-gpu-metrics-exporter",
+    "amd.com/gpu.deploy.vfio-manager": "amd-vfio-manager",
+    "amd.com/gpu.deploy.test-runner": "amd-gpu-test-runner",
+}
+
+
+def pause_value(value: Optional[str]) -> str:
+    """Paused form of a component label value.
+
+    '' / missing and 'false' mean "user disabled" and pass through;
+    'true' becomes the paused marker; an already-paused value is stable
+    (idempotent); any custom value gets a reversible '_<marker>' suffix.
+    """
+    if not value:
+        return ""
+    if value == "false":
+        return "false"
+    if value == "true":
+        return PAUSED_VALUE
+    if PAUSED_VALUE in value:
+        return value
+    return value + "_" + PAUSED_VALUE
+
+
+def unpause_value(value: Optional[str]) -> str:
+    """Inverse of :func:`pause_value` (derivable from the label alone, so
+    a crash mid-eviction stays recoverable — SURVEY.md §5 checkpoint
+    note)."""
+    if value == "false":
+        return "false"
+    if value == PAUSED_VALUE:
+        return "true"
+    if value and PAUSED_VALUE in value:
+        stripped = value.replace("_" + PAUSED_VALUE, "").replace(PAUSED_VALUE, "")
+        return stripped.strip("_")
+    return value or ""
+
+
+def fetch_component_labels(k8s: K8sClient, node_name: str) -> Dict[str, str]:
+    """Snapshot the current component label values (the restore set)."""
+    node = k8s.get_node(node_name)
+    labels = (node.get("metadata") or {}).get("labels") or {}
+    snapshot = {name: labels.get(name, "") for name in COMPONENT_LABELS}
+    for name, value in snapshot.items():
+        logger.info("component label %s=%r", name, value)
+    return snapshot
+
+
+def evict_components(
+    k8s: K8sClient,
+    node_name: str,
+    operator_namespace: str,
+    current_labels: Dict[str, str],
+    timeout: float = 300.0,
+    poll_interval: float = 2.0,
+) -> bool:
+    """Pause component labels and wait for their pods to drain.
+
+    Returns True when every deployed component's pods are gone (or the
+    drain deadline passed — drain timeout is non-fatal, matching the
+    reference envelope g_o_e.py:205-207). Returns False after UNWINDING
+    the labels if the API rejects the pause patch.
+    """
+    paused = {name: pause_value(v) for name, v in current_labels.items()}
+    try:
+        k8s.patch_node_labels(node_name, paused)
+    except ApiError as e:
+        logger.error("failed to pause component labels: %s", e)
+        return False
+    logger.info("paused %d component labels", len(paused))
+
+    # Components that actually had pods to drain: deployed (non-empty,
+    # non-'false') values only.
+    pending = {
+        COMPONENT_APP_LABELS[name]
+        for name, v in current_labels.items()
+        if v and v != "false" and name in COMPONENT_APP_LABELS
+    }
+    deadline = time.monotonic() + timeout
+    while pending and time.monotonic() < deadline:
+        for app in sorted(pending):
+            try:
+                pods = k8s.list_pods(
+                    operator_namespace,
+                    field_selector=f"spec.nodeName={node_name}",
+                    label_selector=f"app={app}",
+                )
+                n = len(pods.get("items") or [])
+            except ApiError as e:
+                logger.warning("pod-drain poll error for %s: %s", app, e)
+                continue
+            if n == 0:
+                logger.info("%s drained", app)
+                pending.discard(app)
+            else:
+                logger.debug("%s: %d pod(s) remaining", app, n)
+        if pending:
+            time.sleep(poll_interval)
+
+    if pending:
+        logger.warning("drain deadline passed with pods remaining: %s", sorted(pending))
+    return True
+
+
+def reschedule_components(
+    k8s: K8sClient, node_name: str, original_labels: Dict[str, str]
+) -> bool:
+    """Restore component labels so the operator reschedules the pods."""
+    restored = {name: unpause_value(v) for name, v in original_labels.items()}
+    try:
+        k8s.patch_node_labels(node_name, restored)
+    except ApiError as e:
+        logger.error("failed to restore component labels: %s", e)
+        return False
+    logger.info("restored %d component labels", len(restored))
+    return True
+
+
+def unwind_paused_labels(
+    k8s: K8sClient, node_name: str, original_labels: Dict[str, str]
+) -> None:
+    """Best-effort restore after a failed eviction (reference gap:
+    main.py:558-566 leaves components paused on that path)."""
+    try:
+        reschedule_components(k8s, node_name, original_labels)
+    except Exception as e:  # pragma: no cover - double fault
+        logger.error("could not unwind paused labels: %s", e)
+
+
+def set_cc_state_label(k8s: K8sClient, node_name: str, state: str) -> bool:
+    """Publish mode.state + derived ready.state (reference semantics,
+    gpu_operator_eviction.py:262-295)."""
+    ready = ready_value_for_state(state)
+    try:
+        k8s.patch_node_labels(
+            node_name, {CC_STATE_LABEL: state, CC_READY_LABEL: ready}
+        )
+    except ApiError as e:
+        logger.error("failed to set state labels: %s", e)
+        return False
+    logger.info("%s=%s %s=%s", CC_STATE_LABEL, state, CC_READY_LABEL, ready)
+    return True
+
+
+def cordon(k8s: K8sClient, node_name: str) -> bool:
+    try:
+        k8s.set_node_unschedulable(node_name, True)
+        logger.info("cordoned node %s", node_name)
+        return True
+    except ApiError as e:
+        logger.error("cordon failed: %s", e)
+        return False
+
+
+def uncordon(k8s: K8sClient, node_name: str) -> bool:
+    try:
+        k8s.set_node_unschedulable(node_name, False)
+        logger.info("uncordoned node %s", node_name)
+        return True
+    except ApiError as e:
+        logger.error("uncordon failed: %s", e)
+        return False

@@ -1,0 +1,374 @@
+"""In-process fake Kubernetes API server + GPU-operator simulator.
+
+The reference has no test backend at all and recommends testing against
+a live cluster (/root/reference/README_PYTHON.md:57). This module gives
+the CPU-only equivalent of BASELINE.json config 1 (dry-run reconcile on
+kind) without kind: a real HTTP server speaking the exact REST surface
+:mod:`k8s_cc_manager_amd.k8s.client` uses — node get/patch, pod list,
+chunked node watch with resourceVersion semantics, ERROR/410 compaction
+events — plus an operator simulator that deletes component pods when
+their labels are paused and reschedules them on restore (what the real
+GPU operator does in response to gpu_operator_eviction-style labels).
+
+Used by the unit/integration tests and by ``bench.py`` (the eviction +
+watch legs of the measured reconcile are real HTTP round-trips).
+"""
+
+from __future__ import annotations
+
+import json
+import logging
+import threading
+import time
+from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
+from typing import Any, Dict, List, Optional, Tuple
+from urllib.parse import parse_qs, urlparse
+
+from .eviction import COMPONENT_APP_LABELS, PAUSED_VALUE
+
+logger = logging.getLogger(__name__)
+
+
+def _merge_patch(target: Dict[str, Any], patch: Dict[str, Any]) -> None:
+    """JSON merge-patch semantics: dicts merge recursively, None deletes."""
+    for key, value in patch.items():
+        if value is None:
+            target.pop(key, None)
+        elif isinstance(value, dict):
+            sub = target.setdefault(key, {})
+            if not isinstance(sub, dict):
+                target[key] = sub = {}
+            _merge_patch(sub, value)
+        else:
+            target[key] = value
+
+
+class FakeCluster:
+    """State store + HTTP server + operator simulator."""
+
+    def __init__(
+        self,
+        operator_namespace: str = "amd-gpu-operator",
+        schedule_delay: float = 0.0,
+        delete_delay: float = 0.0,
+        operator_tick: float = 0.02,
+    ):
+        self.operator_namespace = operator_namespace
+        self.schedule_delay = schedule_delay
+        self.delete_delay = delete_delay
+        self._operator_tick = operator_tick
+
+        self._lock = threading.Condition()
+        self._rv = 0
+        self._compacted_rv = 0
+        self._nodes: Dict[str, Dict[str, Any]] = {}
+        self._pods: Dict[Tuple[str, str, str], Dict[str, Any]] = {}
+        # (node, app) -> due time for pending operator actions
+        self._pending_create: Dict[Tuple[str, str], float] = {}
+        self._pending_delete: Dict[Tuple[str, str], float] = {}
+        self._events: List[Dict[str, Any]] = []  # {"rv", "type", "node"}
+
+        self._server: Optional[ThreadingHTTPServer] = None
+        self._server_thread: Optional[threading.Thread] = None
+        self._operator_thread: Optional[threading.Thread] = None
+        self._stopping = False
+
+    # ------------------------------------------------------------------
+    # state manipulation (thread-safe)
+    # ------------------------------------------------------------------
+    def add_node(self, name: str, labels: Optional[Dict[str, str]] = None) -> None:
+        with self._lock:
+            self._rv += 1
+            self._nodes[name] = {
+                "kind": "Node",
+                "apiVersion": "v1",
+                "metadata": {
+                    "name": name,
+                    "labels": dict(labels or {}),
+                    "resourceVersion": str(self._rv),
+                },
+                "spec": {},
+                "status": {},
+            }
+            self._record_event("ADDED", name)
+
+    def set_node_label(self, name: str, key: str, value: Optional[str]) -> None:
+        with self._lock:
+            node = self._nodes[name]
+            if value is None:
+                node["metadata"]["labels"].pop(key, None)
+            else:
+                node["metadata"]["labels"][key] = value
+            self._bump(name)
+
+    def get_node_copy(self, name: str) -> Dict[str, Any]:
+        with self._lock:
+            return json.loads(json.dumps(self._nodes[name]))
+
+    def node_labels(self, name: str) -> Dict[str, str]:
+        with self._lock:
+            return dict(self._nodes[name]["metadata"]["labels"])
+
+    def node_unschedulable(self, name: str) -> bool:
+        with self._lock:
+            return bool(self._nodes[name]["spec"].get("unschedulable"))
+
+    def add_pod(self, namespace: str, name: str, node: str, app: str) -> None:
+        with self._lock:
+            self._pods[(namespace, name, node)] = {
+                "kind": "Pod",
+                "apiVersion": "v1",
+                "metadata": {
+                    "name": name,
+                    "namespace": namespace,
+                    "labels": {"app": app},
+                },
+                "spec": {"nodeName": node},
+                "status": {"phase": "Running"},
+            }
+
+    def pods_on(self, node: str, app: Optional[str] = None) -> List[Dict[str, Any]]:
+        with self._lock:
+            out = []
+            for pod in self._pods.values():
+                if pod["spec"]["nodeName"] != node:
+                    continue
+                if app and pod["metadata"]["labels"].get("app") != app:
+                    continue
+                out.append(json.loads(json.dumps(pod)))
+            return out
+
+    def compact(self) -> None:
+        """Mark all current events compacted -> old-RV watches get 410."""
+        with self._lock:
+            self._compacted_rv = self._rv
+            self._events = [e for e in self._events if e["rv"] > self._compacted_rv]
+
+    def _bump(self, name: str) -> None:
+        self._rv += 1
+        self._nodes[name]["metadata"]["resourceVersion"] = str(self._rv)
+        self._record_event("MODIFIED", name)
+
+    def _record_event(self, etype: str, name: str) -> None:
+        self._events.append(
+            {
+                "rv": self._rv,
+                "type": etype,
+                "node": json.loads(json.dumps(self._nodes[name])),
+            }
+        )
+        if len(self._events) > 4096:
+            self._events = self._events[-2048:]
+            self._compacted_rv = max(self._compacted_rv, self._events[0]["rv"] - 1)
+        self._lock.notify_all()
+
+    # ------------------------------------------------------------------
+    # operator simulator
+    # ------------------------------------------------------------------
+    def _operator_loop(self) -> None:
+        while not self._stopping:
+            now = time.monotonic()
+            with self._lock:
+                for node_name, node in self._nodes.items():
+                    labels = node["metadata"]["labels"]
+                    for comp_label, app in COMPONENT_APP_LABELS.items():
+                        value = labels.get(comp_label, "")
+                        deployed = bool(value) and value != "false" and PAUSED_VALUE not in value
+                        key = (node_name, app)
+                        pod_key = (self.operator_namespace, f"{app}-{node_name}", node_name)
+                        exists = pod_key in self._pods
+                        if deployed and not exists:
+                            due = self._pending_create.setdefault(
+                                key, now + self.schedule_delay
+                            )
+                            if now >= due:
+                                self._pods[pod_key] = {
+                                    "kind": "Pod",
+                                    "apiVersion": "v1",
+                                    "metadata": {
+                                        "name": pod_key[1],
+                                        "namespace": self.operator_namespace,
+                                        "labels": {"app": app},
+                                    },
+                                    "spec": {"nodeName": node_name},
+                                    "status": {"phase": "Running"},
+                                }
+                                self._pending_create.pop(key, None)
+                        elif not deployed and exists:
+                            due = self._pending_delete.setdefault(
+                                key, now + self.delete_delay
+                            )
+                            if now >= due:
+                                self._pods.pop(pod_key, None)
+                                self._pending_delete.pop(key, None)
+                        else:
+                            self._pending_create.pop(key, None)
+                            self._pending_delete.pop(key, None)
+            time.sleep(self._operator_tick)
+
+    # ------------------------------------------------------------------
+    # HTTP server
+    # ------------------------------------------------------------------
+    def start(self) -> str:
+        cluster = self
+
+        class Handler(BaseHTTPRequestHandler):
+            protocol_version = "HTTP/1.1"
+
+            def log_message(self, fmt: str, *args: Any) -> None:  # quiet
+                logger.debug("fakeapi: " + fmt, *args)
+
+            def _send_json(self, code: int, obj: Dict[str, Any]) -> None:
+                body = json.dumps(obj).encode()
+                self.send_response(code)
+                self.send_header("Content-Type", "application/json")
+                self.send_header("Content-Length", str(len(body)))
+                self.end_headers()
+                self.wfile.write(body)
+
+            def do_GET(self) -> None:
+                url = urlparse(self.path)
+                qs = parse_qs(url.query)
+                parts = [p for p in url.path.split("/") if p]
+                try:
+                    if parts[:3] == ["api", "v1", "nodes"] and len(parts) == 4:
+                        name = parts[3]
+                        with cluster._lock:
+                            node = cluster._nodes.get(name)
+                            if node is None:
+                                return self._send_json(
+                                    404, {"kind": "Status", "code": 404}
+                                )
+                            return self._send_json(200, json.loads(json.dumps(node)))
+                    if parts[:3] == ["api", "v1", "nodes"] and qs.get("watch"):
+                        return self._watch_nodes(qs)
+                    if (
+                        len(parts) == 5
+                        and parts[:3] == ["api", "v1", "namespaces"]
+                        and parts[4] == "pods"
+                    ):
+                        return self._list_pods(parts[3], qs)
+                    self._send_json(404, {"kind": "Status", "code": 404})
+                except (BrokenPipeError, ConnectionResetError):
+                    pass
+
+            def do_PATCH(self) -> None:
+                url = urlparse(self.path)
+                parts = [p for p in url.path.split("/") if p]
+                length = int(self.headers.get("Content-Length", 0))
+                patch = json.loads(self.rfile.read(length) or b"{}")
+                if parts[:3] == ["api", "v1", "nodes"] and len(parts) == 4:
+                    name = parts[3]
+                    with cluster._lock:
+                        node = cluster._nodes.get(name)
+                        if node is None:
+                            return self._send_json(404, {"kind": "Status", "code": 404})
+                        _merge_patch(node, patch)
+                        cluster._bump(name)
+                        return self._send_json(200, json.loads(json.dumps(node)))
+                self._send_json(404, {"kind": "Status", "code": 404})
+
+            # -- helpers ------------------------------------------------
+            def _list_pods(self, namespace: str, qs: Dict[str, List[str]]) -> None:
+                field_sel = (qs.get("fieldSelector") or [""])[0]
+                label_sel = (qs.get("labelSelector") or [""])[0]
+                want_node = None
+                if field_sel.startswith("spec.nodeName="):
+                    want_node = field_sel.split("=", 1)[1]
+                want_app = None
+                if label_sel.startswith("app="):
+                    want_app = label_sel.split("=", 1)[1]
+                with cluster._lock:
+                    items = []
+                    for (ns, _pname, node), pod in cluster._pods.items():
+                        if ns != namespace:
+                            continue
+                        if want_node and node != want_node:
+                            continue
+                        if want_app and pod["metadata"]["labels"].get("app") != want_app:
+                            continue
+                        items.append(json.loads(json.dumps(pod)))
+                self._send_json(200, {"kind": "PodList", "items": items})
+
+            def _watch_nodes(self, qs: Dict[str, List[str]]) -> None:
+                field_sel = (qs.get("fieldSelector") or [""])[0]
+                want = field_sel.split("=", 1)[1] if "=" in field_sel else None
+                rv = int((qs.get("resourceVersion") or ["0"])[0] or "0")
+                timeout = float((qs.get("timeoutSeconds") or ["300"])[0])
+                deadline = time.monotonic() + timeout
+
+                self.send_response(200)
+                self.send_header("Content-Type", "application/json")
+                self.send_header("Transfer-Encoding", "chunked")
+                self.end_headers()
+
+                def send_chunk(obj: Dict[str, Any]) -> None:
+                    data = (json.dumps(obj) + "\n").encode()
+                    self.wfile.write(f"{len(data):x}\r\n".encode() + data + b"\r\n")
+                    self.wfile.flush()
+
+                with cluster._lock:
+                    if rv and rv < cluster._compacted_rv:
+                        send_chunk(
+                            {
+                                "type": "ERROR",
+                                "object": {
+                                    "kind": "Status",
+                                    "code": 410,
+                                    "reason": "Expired",
+                                    "message": f"too old resource version: {rv}",
+                                },
+                            }
+                        )
+                        self.wfile.write(b"0\r\n\r\n")
+                        return
+                last_sent = rv
+                try:
+                    while time.monotonic() < deadline:
+                        with cluster._lock:
+                            pending = [
+                                e
+                                for e in cluster._events
+                                if e["rv"] > last_sent
+                                and (not want or e["node"]["metadata"]["name"] == want)
+                            ]
+                            if not pending:
+                                cluster._lock.wait(
+                                    timeout=min(0.25, max(0.0, deadline - time.monotonic()))
+                                )
+                                pending = [
+                                    e
+                                    for e in cluster._events
+                                    if e["rv"] > last_sent
+                                    and (
+                                        not want
+                                        or e["node"]["metadata"]["name"] == want
+                                    )
+                                ]
+                        for event in pending:
+                            send_chunk({"type": event["type"], "object": event["node"]})
+                            last_sent = event["rv"]
+                    self.wfile.write(b"0\r\n\r\n")
+                except (BrokenPipeError, ConnectionResetError):
+                    pass
+
+        self._server = ThreadingHTTPServer(("127.0.0.1", 0), Handler)
+        self._server.daemon_threads = True
+        self._server_thread = threading.Thread(
+            target=self._server.serve_forever, name="fake-apiserver", daemon=True
+        )
+        self._server_thread.start()
+        self._operator_thread = threading.Thread(
+            target=self._operator_loop, name="fake-operator", daemon=True
+        )
+        self._operator_thread.start()
+        host, port = self._server.server_address
+        return f"http://{host}:{port}"
+
+    def stop(self) -> None:
+        self._stopping = True
+        if self._server:
+            self._server.shutdown()
+            self._server.server_close()
+        if self._operator_thread:
+            self._operator_thread.join(timeout=2)

@@ -1,0 +1,43 @@
+"""Node-label API of the AMD CC manager.
+
+Shape-compatible with the reference's label contract
+(/root/reference/main.py:62, /root/reference/gpu_operator_eviction.py:262-295)
+but in the ``amd.com`` namespace:
+
+- input   ``amd.com/gpu.cc.mode``        in {on, off, devtools, ppcie}
+- output  ``amd.com/gpu.cc.mode.state``  = the applied mode, or "failed"
+- output  ``amd.com/gpu.cc.ready.state`` = "true" | "false" | ""
+
+``ppcie`` selects the fabric-protected (xGMI-hive-wide TEE-IO) mode, the
+AMD analogue of the reference's Protected-PCIe multi-GPU mode.
+"""
+
+CC_MODE_LABEL = "amd.com/gpu.cc.mode"
+CC_STATE_LABEL = "amd.com/gpu.cc.mode.state"
+CC_READY_LABEL = "amd.com/gpu.cc.ready.state"
+
+MODE_ON = "on"
+MODE_OFF = "off"
+MODE_DEVTOOLS = "devtools"
+MODE_PPCIE = "ppcie"  # fabric-protected (xGMI hive) mode
+STATE_FAILED = "failed"
+
+VALID_MODES = (MODE_ON, MODE_OFF, MODE_DEVTOOLS, MODE_PPCIE)
+
+# Modes in which workloads may attest into the GPU TEE
+# (reference semantics: gpu_operator_eviction.py:276-279).
+READY_MODES = (MODE_ON, MODE_PPCIE)
+
+
+def ready_value_for_state(state: str) -> str:
+    """Derive ``cc.ready.state`` from ``cc.mode.state``.
+
+    on/ppcie -> "true"; off -> "false"; anything else (devtools, failed,
+    empty) -> "" — same derivation the reference applies at
+    gpu_operator_eviction.py:275-279.
+    """
+    if state in READY_MODES:
+        return "true"
+    if state == MODE_OFF:
+        return "false"
+    return ""

@@ -1,0 +1,130 @@
+#!/usr/bin/env python3
+"""Process entrypoint of the AMD CC manager DaemonSet container.
+
+Flag/env surface mirrors the reference (/root/reference/main.py:698-763):
+``--kubeconfig``, ``--default-cc-mode/-m``, ``--node-name``, ``--debug``
+plus env NODE_NAME, DEFAULT_CC_MODE, OPERATOR_NAMESPACE,
+EVICT_OPERATOR_COMPONENTS, CORDON_NODE, CC_READINESS_FILE, KUBECONFIG,
+CC_DEVICE_BACKEND (mock|amdsmi|auto), CC_METRICS_PORT.
+"""
+
+from __future__ import annotations
+
+import argparse
+import os
+import sys
+
+from .core.hostprobe import is_host_cc_enabled
+from .core.manager import CCManager, FatalConfigError, ManagerConfig
+from .core.transition import TransitionEngine
+from .device import get_backend
+from .k8s.client import load_client
+from .labels import MODE_OFF
+from .utils.logging import setup_logging
+from .utils.metrics import METRICS
+
+
+def build_parser() -> argparse.ArgumentParser:
+    parser = argparse.ArgumentParser(
+        prog="cc-manager-amd",
+        description="AMD MI355X Confidential-Computing manager for Kubernetes",
+    )
+    parser.add_argument(
+        "--kubeconfig",
+        default=os.environ.get("KUBECONFIG", ""),
+        help="path to kubeconfig (default: in-cluster config)",
+    )
+    parser.add_argument(
+        "--default-cc-mode",
+        "-m",
+        default=os.environ.get("DEFAULT_CC_MODE", "on"),
+        help="mode applied when the amd.com/gpu.cc.mode label is absent "
+        "(on|off|devtools|ppcie)",
+    )
+    parser.add_argument(
+        "--node-name",
+        default=os.environ.get("NODE_NAME", ""),
+        help="Kubernetes node name (default: $NODE_NAME)",
+    )
+    parser.add_argument(
+        "--device-backend",
+        default=os.environ.get("CC_DEVICE_BACKEND", "auto"),
+        choices=["auto", "amdsmi", "mock"],
+        help="device layer implementation",
+    )
+    parser.add_argument(
+        "--metrics-port",
+        type=int,
+        default=int(os.environ.get("CC_METRICS_PORT", "0")),
+        help="serve Prometheus metrics on this port (0 = off)",
+    )
+    parser.add_argument("--debug", action="store_true", help="debug logging")
+    return parser
+
+
+def main(argv=None) -> int:
+    args = build_parser().parse_args(argv)
+    logger = setup_logging(args.debug)
+
+    if not args.node_name:
+        logger.error("NODE_NAME must be set")
+        return 1
+
+    default_mode = args.default_cc_mode
+    host_cc = is_host_cc_enabled()
+    if not host_cc and default_mode != MODE_OFF:
+        logger.warning(
+            "host does not support CC; overriding default mode %r -> 'off'",
+            default_mode,
+        )
+        default_mode = MODE_OFF
+
+    if args.metrics_port:
+        METRICS.serve(args.metrics_port)
+
+    try:
+        k8s = load_client(args.kubeconfig)
+    except Exception as e:
+        logger.error("failed to load Kubernetes configuration: %s", e)
+        return 1
+
+    backend = get_backend(args.device_backend)
+    engine = TransitionEngine(attestor=_maybe_attestor(logger))
+    manager = CCManager(
+        node_name=args.node_name,
+        default_mode=default_mode,
+        host_cc=host_cc,
+        k8s=k8s,
+        backend=backend,
+        engine=engine,
+        config=ManagerConfig.from_env(),
+    )
+    try:
+        manager.run()
+        return 0
+    except KeyboardInterrupt:
+        logger.info("shutting down")
+        return 0
+    except FatalConfigError as e:
+        logger.error("fatal node configuration error: %s", e)
+        return 1
+    except Exception as e:
+        logger.error("fatal error: %s", e, exc_info=True)
+        return 1
+
+
+def _maybe_attestor(logger):
+    """Load the HIP attestation probe when a GPU is present; None on
+    CPU-only nodes (mock runs)."""
+    try:
+        from .ops.attest import attest_device_by_bdf, probe_available
+
+        if probe_available():
+            return attest_device_by_bdf
+    except Exception as e:  # pragma: no cover
+        logger.warning("attestation probe unavailable: %s", e)
+    return None
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -1,0 +1,91 @@
+"""The minimal K8s client against the fake API server (real HTTP)."""
+
+import threading
+import time
+
+import pytest
+
+from k8s_cc_manager_amd.k8s.client import ApiError, K8sClient
+
+
+@pytest.fixture
+def cluster_client(fake_cluster):
+    cluster, url = fake_cluster
+    cluster.add_node("node0", labels={"a": "1"})
+    return cluster, K8sClient(url)
+
+
+def test_get_node(cluster_client):
+    cluster, k8s = cluster_client
+    node = k8s.get_node("node0")
+    assert node["metadata"]["name"] == "node0"
+    assert node["metadata"]["labels"]["a"] == "1"
+
+
+def test_get_missing_node_raises(cluster_client):
+    _, k8s = cluster_client
+    with pytest.raises(ApiError) as ei:
+        k8s.get_node("ghost")
+    assert ei.value.status == 404
+
+
+def test_patch_labels_merge_and_delete(cluster_client):
+    cluster, k8s = cluster_client
+    k8s.patch_node_labels("node0", {"b": "2"})
+    assert cluster.node_labels("node0") == {"a": "1", "b": "2"}
+    k8s.patch_node_labels("node0", {"a": None})
+    assert cluster.node_labels("node0") == {"b": "2"}
+
+
+def test_cordon_uncordon(cluster_client):
+    cluster, k8s = cluster_client
+    k8s.set_node_unschedulable("node0", True)
+    assert cluster.node_unschedulable("node0")
+    k8s.set_node_unschedulable("node0", False)
+    assert not cluster.node_unschedulable("node0")
+
+
+def test_list_pods_selectors(cluster_client):
+    cluster, k8s = cluster_client
+    cluster.add_pod("ns1", "p1", "node0", app="appA")
+    cluster.add_pod("ns1", "p2", "node0", app="appB")
+    cluster.add_pod("ns1", "p3", "other", app="appA")
+    pods = k8s.list_pods("ns1", field_selector="spec.nodeName=node0",
+                         label_selector="app=appA")
+    names = [p["metadata"]["name"] for p in pods["items"]]
+    assert names == ["p1"]
+
+
+def test_watch_sees_label_change(cluster_client):
+    cluster, k8s = cluster_client
+    node = k8s.get_node("node0")
+    rv = node["metadata"]["resourceVersion"]
+
+    seen = []
+
+    def mutate():
+        time.sleep(0.1)
+        cluster.set_node_label("node0", "amd.com/gpu.cc.mode", "on")
+
+    t = threading.Thread(target=mutate)
+    t.start()
+    for event in k8s.watch_node("node0", resource_version=rv, timeout_seconds=3):
+        seen.append(event)
+        labels = event["object"]["metadata"].get("labels", {})
+        if labels.get("amd.com/gpu.cc.mode") == "on":
+            break
+    t.join()
+    assert any(
+        e["object"]["metadata"]["labels"].get("amd.com/gpu.cc.mode") == "on"
+        for e in seen
+    )
+
+
+def test_watch_410_on_compacted_rv(cluster_client):
+    cluster, k8s = cluster_client
+    for i in range(5):
+        cluster.set_node_label("node0", "x", str(i))
+    cluster.compact()
+    events = list(k8s.watch_node("node0", resource_version="1", timeout_seconds=2))
+    assert events and events[0]["type"] == "ERROR"
+    assert events[0]["object"]["code"] == 410

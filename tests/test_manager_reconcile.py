@@ -1,0 +1,229 @@
+"""End-to-end reconcile against the fake API server + mock devices —
+the CPU-only equivalent of BASELINE.json config 1."""
+
+import threading
+import time
+
+import pytest
+
+from k8s_cc_manager_amd.core.manager import CCManager, FatalConfigError, ManagerConfig
+from k8s_cc_manager_amd.core.transition import TransitionEngine
+from k8s_cc_manager_amd.device.mock import FaultPlan, MockBackend
+from k8s_cc_manager_amd.k8s.client import K8sClient
+from k8s_cc_manager_amd.k8s.eviction import COMPONENT_APP_LABELS, COMPONENT_LABELS
+from k8s_cc_manager_amd.labels import (
+    CC_MODE_LABEL,
+    CC_READY_LABEL,
+    CC_STATE_LABEL,
+)
+
+NODE = "node0"
+
+
+def make_manager(cluster, url, backend=None, evict=True, cordon=True, **cfg_kw):
+    cfg = ManagerConfig(
+        evict_components=evict,
+        cordon_node=cordon,
+        eviction_timeout=5.0,
+        eviction_poll_interval=0.05,
+        reconnect_backoff=0.1,
+        watch_timeout_seconds=2,
+        **cfg_kw,
+    )
+    return CCManager(
+        node_name=NODE,
+        default_mode="on",
+        host_cc=True,
+        k8s=K8sClient(url),
+        backend=backend or MockBackend(num_gpus=8),
+        engine=TransitionEngine(),
+        config=cfg,
+    )
+
+
+@pytest.fixture
+def node_cluster(fake_cluster):
+    cluster, url = fake_cluster
+    labels = {name: "true" for name in COMPONENT_LABELS}
+    cluster.add_node(NODE, labels=labels)
+    # let the operator simulator schedule the component pods
+    deadline = time.monotonic() + 2
+    while time.monotonic() < deadline:
+        if len(cluster.pods_on(NODE)) == len(COMPONENT_APP_LABELS):
+            break
+        time.sleep(0.02)
+    assert len(cluster.pods_on(NODE)) == len(COMPONENT_APP_LABELS)
+    return cluster, url
+
+
+def test_full_reconcile_on(node_cluster):
+    cluster, url = node_cluster
+    mgr = make_manager(cluster, url)
+    assert mgr.apply_mode("on")
+
+    labels = cluster.node_labels(NODE)
+    assert labels[CC_STATE_LABEL] == "on"
+    assert labels[CC_READY_LABEL] == "true"
+    # components restored -> pods back
+    deadline = time.monotonic() + 2
+    while time.monotonic() < deadline:
+        if len(cluster.pods_on(NODE)) == len(COMPONENT_APP_LABELS):
+            break
+        time.sleep(0.02)
+    assert len(cluster.pods_on(NODE)) == len(COMPONENT_APP_LABELS)
+    # node uncordoned at the end
+    assert not cluster.node_unschedulable(NODE)
+    # component labels restored to originals
+    for name in COMPONENT_LABELS:
+        assert cluster.node_labels(NODE)[name] == "true"
+
+
+def test_pods_drained_before_device_work(node_cluster):
+    cluster, url = node_cluster
+    backend = MockBackend(num_gpus=2)
+    observed = {}
+
+    class SpyEngine(TransitionEngine):
+        def apply_cc_mode(self, all_devices, gpus, mode):
+            observed["pods_at_transition"] = len(cluster.pods_on(NODE))
+            observed["cordoned"] = cluster.node_unschedulable(NODE)
+            return super().apply_cc_mode(all_devices, gpus, mode)
+
+    mgr = make_manager(cluster, url, backend=backend)
+    mgr.engine = SpyEngine()
+    assert mgr.apply_mode("on")
+    assert observed["pods_at_transition"] == 0  # fully drained first
+    assert observed["cordoned"] is True
+
+
+def test_idempotent_apply_skips_eviction(node_cluster):
+    cluster, url = node_cluster
+    backend = MockBackend(num_gpus=4, initial_cc_mode="on")
+    mgr = make_manager(cluster, url, backend=backend)
+    assert mgr.apply_mode("on")
+    # no eviction: component labels never paused, pods still there
+    assert len(cluster.pods_on(NODE)) == len(COMPONENT_APP_LABELS)
+    assert cluster.node_labels(NODE)[CC_STATE_LABEL] == "on"
+
+
+def test_failed_transition_labels_failed(node_cluster):
+    cluster, url = node_cluster
+    backend = MockBackend(num_gpus=4, faults=FaultPlan(fail_reset=["0000:18:00.0"]))
+    mgr = make_manager(cluster, url, backend=backend)
+    assert not mgr.apply_mode("on")
+    labels = cluster.node_labels(NODE)
+    assert labels[CC_STATE_LABEL] == "failed"
+    assert labels[CC_READY_LABEL] == ""
+    # components still rescheduled and node uncordoned after failure
+    for name in COMPONENT_LABELS:
+        assert labels[name] == "true"
+    assert not cluster.node_unschedulable(NODE)
+
+
+def test_mixed_capability_is_fatal(node_cluster):
+    cluster, url = node_cluster
+    backend = MockBackend(num_gpus=4)
+    backend.device(2)._cc_capable = False
+    mgr = make_manager(cluster, url, backend=backend)
+    with pytest.raises(FatalConfigError):
+        mgr.apply_mode("on")
+
+
+def test_off_mode_with_partial_capability_ok(node_cluster):
+    cluster, url = node_cluster
+    backend = MockBackend(num_gpus=4, initial_cc_mode="on")
+    backend.device(2)._cc_capable = False
+    backend.device(2)._cc_mode = "off"
+    mgr = make_manager(cluster, url, backend=backend)
+    assert mgr.apply_mode("off")
+    assert cluster.node_labels(NODE)[CC_STATE_LABEL] == "off"
+
+
+def test_no_gpus_is_noop_true(node_cluster):
+    cluster, url = node_cluster
+    mgr = make_manager(cluster, url, backend=MockBackend(num_gpus=0))
+    assert mgr.apply_mode("on")
+
+
+def test_invalid_mode_labels_failed(node_cluster):
+    cluster, url = node_cluster
+    mgr = make_manager(cluster, url)
+    assert not mgr.apply_mode("bogus")
+    assert cluster.node_labels(NODE)[CC_STATE_LABEL] == "failed"
+
+
+def test_ppcie_reconcile(node_cluster):
+    cluster, url = node_cluster
+    backend = MockBackend(num_gpus=4)
+    mgr = make_manager(cluster, url, backend=backend)
+    assert mgr.apply_mode("ppcie")
+    labels = cluster.node_labels(NODE)
+    assert labels[CC_STATE_LABEL] == "ppcie"
+    assert labels[CC_READY_LABEL] == "true"
+    assert all(d.query_fabric_mode() == "on" for d in backend.get_gpus())
+
+
+def test_devtools_mode_ready_empty(node_cluster):
+    cluster, url = node_cluster
+    mgr = make_manager(cluster, url)
+    assert mgr.apply_mode("devtools")
+    labels = cluster.node_labels(NODE)
+    assert labels[CC_STATE_LABEL] == "devtools"
+    assert labels[CC_READY_LABEL] == ""
+
+
+def test_watch_loop_applies_label_change(node_cluster):
+    cluster, url = node_cluster
+    backend = MockBackend(num_gpus=2)
+    mgr = make_manager(cluster, url, backend=backend)
+    cluster.set_node_label(NODE, CC_MODE_LABEL, "off")
+
+    t = threading.Thread(target=mgr.run, daemon=True)
+    t.start()
+    # wait for initial apply (label 'off')
+    deadline = time.monotonic() + 5
+    while time.monotonic() < deadline:
+        if cluster.node_labels(NODE).get(CC_STATE_LABEL) == "off":
+            break
+        time.sleep(0.02)
+    assert cluster.node_labels(NODE).get(CC_STATE_LABEL) == "off"
+
+    # flip desired mode -> watch must apply it
+    cluster.set_node_label(NODE, CC_MODE_LABEL, "on")
+    deadline = time.monotonic() + 5
+    while time.monotonic() < deadline:
+        if cluster.node_labels(NODE).get(CC_STATE_LABEL) == "on":
+            break
+        time.sleep(0.02)
+    assert cluster.node_labels(NODE).get(CC_STATE_LABEL) == "on"
+    assert all(m == "on" for m in backend.modes().values())
+
+    mgr.stop_event.set()
+    t.join(timeout=5)
+
+
+def test_watch_resyncs_after_compaction(node_cluster):
+    cluster, url = node_cluster
+    backend = MockBackend(num_gpus=1)
+    mgr = make_manager(cluster, url, backend=backend)
+    cluster.set_node_label(NODE, CC_MODE_LABEL, "off")
+    t = threading.Thread(target=mgr.run, daemon=True)
+    t.start()
+    deadline = time.monotonic() + 5
+    while time.monotonic() < deadline:
+        if cluster.node_labels(NODE).get(CC_STATE_LABEL) == "off":
+            break
+        time.sleep(0.02)
+
+    # compact event history so the manager's RV turns stale, then change
+    # the label: the 410 path must resync and apply.
+    cluster.set_node_label(NODE, CC_MODE_LABEL, "devtools")
+    cluster.compact()
+    deadline = time.monotonic() + 8
+    while time.monotonic() < deadline:
+        if cluster.node_labels(NODE).get(CC_STATE_LABEL) == "devtools":
+            break
+        time.sleep(0.02)
+    assert cluster.node_labels(NODE).get(CC_STATE_LABEL) == "devtools"
+    mgr.stop_event.set()
+    t.join(timeout=5)
