@@ -1,0 +1,205 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: CC-mode reconcile transitions on MI355X.
+
+Measures BASELINE.json's metric — CC-mode transition sec/GPU and
+reconcile GPUs/sec at 1/2/4/8 GPUs — on a synthetic single-node cluster
+(in-process fake Kubernetes API server with a simulated GPU operator,
+exactly the reference's label/eviction protocol).
+
+One STEP = one full reconcile transition of every managed GPU:
+  desired-mode label flip on the (fake) API server -> read label ->
+  cordon -> evict operator components (pod-drain poll over HTTP) ->
+  4-phase device transition (fabric-off, stage-all, reset-all,
+  boot-wait + verify + HIP attestation probe) -> state labels ->
+  reschedule -> uncordon.
+Steps alternate on -> off -> on, so every step transitions every GPU.
+
+Device tier (reported in config.device_tier):
+- with a GPU: the shadow backend — real enumeration, real post-reset
+  liveness kernel, real MFMA+LDS+HBM attestation probe on the device;
+  only the privileged CC register write itself is shadowed (an actual
+  FLR would kill the shared box; see device/shadow.py).
+- without a GPU (or --mock): the mock backend with a synthetic
+  reset/boot latency envelope.
+
+Scaling: weak (each rank/GPU does a fixed amount of work). Under
+torchrun (one rank per GPU) the stage->reset seam is synchronized
+across ranks with a torch.distributed barrier — the xGMI-hive
+stage-all-then-reset-all invariant.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+
+def parse_args(argv=None):
+    p = argparse.ArgumentParser(description=__doc__)
+    p.add_argument("--gpus", type=int, default=1, help="GPUs to manage (single-process) or ranks (torchrun)")
+    p.add_argument("--steps", type=int, default=10, help="timed reconcile steps")
+    p.add_argument("--warmup", type=int, default=2, help="untimed warmup steps")
+    p.add_argument("--attest-dim", type=int, default=1024, help="attestation GEMM size")
+    p.add_argument("--mock", action="store_true", help="force the mock device tier")
+    p.add_argument("--no-evict", action="store_true", help="skip the eviction leg")
+    p.add_argument("--json-out", default="", help="also write the JSON line here")
+    return p.parse_args(argv)
+
+
+def main(argv=None) -> int:
+    args = parse_args(argv)
+
+    import logging
+
+    logging.basicConfig(level=logging.WARNING)
+
+    import torch
+    import torch.distributed as dist
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+
+    use_gpu = torch.cuda.is_available() and not args.mock
+    if world > 1:
+        backend = "nccl" if use_gpu else "gloo"
+        if use_gpu:
+            torch.cuda.set_device(local_rank)
+        dist.init_process_group(backend=backend)
+
+    from k8s_cc_manager_amd.core.manager import CCManager, ManagerConfig
+    from k8s_cc_manager_amd.core.transition import TransitionEngine
+    from k8s_cc_manager_amd.k8s.client import K8sClient
+    from k8s_cc_manager_amd.k8s.eviction import COMPONENT_LABELS
+    from k8s_cc_manager_amd.k8s.fakecluster import FakeCluster
+    from k8s_cc_manager_amd.labels import CC_MODE_LABEL
+    from k8s_cc_manager_amd.parallel.fabric import DistFabricBarrier
+
+    # ---- per-rank synthetic cluster -----------------------------------
+    node_name = f"bench-node-{rank}"
+    cluster = FakeCluster(operator_tick=0.005)
+    url = cluster.start()
+    cluster.add_node(node_name, labels={name: "true" for name in COMPONENT_LABELS})
+    k8s = K8sClient(url)
+
+    # ---- device tier --------------------------------------------------
+    n_managed = args.gpus if world == 1 else 1
+    if use_gpu:
+        from k8s_cc_manager_amd.device.shadow import ShadowBackend
+        from k8s_cc_manager_amd.ops import attest
+
+        n_visible = torch.cuda.device_count()
+        indices = [local_rank] if world > 1 else [i % n_visible for i in range(n_managed)]
+        backend_dev = ShadowBackend(device_indices=indices)
+        attestor = lambda dev: attest.attest_device(  # noqa: E731
+            dev.hip_index, gemm_dim=args.attest_dim
+        )
+        device_tier = "shadow+hip-attest"
+    else:
+        from k8s_cc_manager_amd.device.mock import MockBackend, MockLatency
+
+        backend_dev = MockBackend(
+            num_gpus=n_managed, latency=MockLatency(reset=0.02, boot=0.05)
+        )
+        attestor = None
+        device_tier = "mock"
+
+    barrier = DistFabricBarrier() if world > 1 else None
+    engine = TransitionEngine(attestor=attestor, barrier=barrier)
+    manager = CCManager(
+        node_name=node_name,
+        default_mode="on",
+        host_cc=True,
+        k8s=k8s,
+        backend=backend_dev,
+        engine=engine,
+        config=ManagerConfig(
+            evict_components=not args.no_evict,
+            cordon_node=True,
+            eviction_timeout=30.0,
+            eviction_poll_interval=0.01,
+        ),
+    )
+
+    def reconcile_step(i: int) -> None:
+        mode = "on" if i % 2 == 0 else "off"
+        cluster.set_node_label(node_name, CC_MODE_LABEL, mode)
+        label = manager.read_mode_label()
+        ok = manager.apply_mode(manager.with_default(label))
+        if not ok:
+            raise RuntimeError(f"rank {rank}: reconcile step {i} failed")
+
+    def sync() -> None:
+        if use_gpu:
+            torch.cuda.synchronize()
+        if world > 1:
+            dist.barrier()
+
+    # ---- warmup -------------------------------------------------------
+    for i in range(args.warmup):
+        reconcile_step(i)
+    sync()
+
+    # ---- timed region -------------------------------------------------
+    t0 = time.perf_counter()
+    for i in range(args.warmup, args.warmup + args.steps):
+        reconcile_step(i)
+    sync()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    if world > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        if use_gpu:
+            t = t.cuda()
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    n_gpus_total = world * n_managed if world > 1 else n_managed
+    gpus_per_sec = n_gpus_total * args.steps / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if rank == 0:
+        result = {
+            "metric": "reconcile_gpus_per_sec",
+            "value": round(gpus_per_sec, 4),
+            "unit": "GPU-transitions/s",
+            "n_gpus": n_gpus_total,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": "cc-mode-transition",
+                "global_batch": n_gpus_total,
+                "seq_len": args.attest_dim,
+                "parallelism": f"concurrent-per-gpu x{n_gpus_total}",
+                "modes": "on<->off toggle",
+                "device_tier": device_tier,
+                "eviction": not args.no_evict,
+                "components": len(COMPONENT_LABELS),
+                "attest_gemm_dim": args.attest_dim if use_gpu else 0,
+                "sec_per_gpu_transition": round(elapsed / args.steps, 4),
+            },
+        }
+        line = json.dumps(result)
+        print(line)
+        if args.json_out:
+            with open(args.json_out, "w") as f:
+                f.write(line + "\n")
+
+    cluster.stop()
+    if world > 1:
+        dist.destroy_process_group()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

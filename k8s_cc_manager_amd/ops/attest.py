@@ -1,0 +1,228 @@
+"""Python surface of the CDNA4 attestation probe.
+
+Loads the in-tree ``libccattest.so`` via ctypes (the control-plane
+daemon must not depend on torch). On a GPU box the library is REQUIRED:
+if AMD GPUs are visible but the library is missing or fails to load,
+attestation raises instead of silently passing — a reset GPU may never
+be labeled ready on the strength of a Python fallback.
+"""
+
+from __future__ import annotations
+
+import ctypes
+import logging
+import os
+import re
+from dataclasses import dataclass
+from pathlib import Path
+from typing import Optional
+
+logger = logging.getLogger(__name__)
+
+_LIB_PATH = Path(__file__).resolve().parent / "libccattest.so"
+_lib: Optional[ctypes.CDLL] = None
+
+
+class AttestationError(Exception):
+    """The post-reset attestation probe failed — the device must NOT be
+    labeled ready."""
+
+
+class _CReport(ctypes.Structure):
+    _fields_ = [
+        ("device", ctypes.c_int),
+        ("cu_count", ctypes.c_int),
+        ("xcc_count", ctypes.c_int),
+        ("arch", ctypes.c_char * 64),
+        ("error", ctypes.c_char * 256),
+        ("vram_total_mb", ctypes.c_longlong),
+        ("gemm_m", ctypes.c_int),
+        ("gemm_n", ctypes.c_int),
+        ("gemm_k", ctypes.c_int),
+        ("gemm_ms", ctypes.c_double),
+        ("gemm_tflops", ctypes.c_double),
+        ("ref_ms", ctypes.c_double),
+        ("max_abs_err", ctypes.c_float),
+        ("checksum", ctypes.c_ulonglong),
+        ("lds_ms", ctypes.c_double),
+        ("lds_failures", ctypes.c_uint),
+        ("hbm_ms", ctypes.c_double),
+        ("hbm_gbps", ctypes.c_double),
+        ("peer_count", ctypes.c_int),
+        ("peers_accessible", ctypes.c_int),
+        ("ok", ctypes.c_int),
+    ]
+
+
+@dataclass
+class AttestReport:
+    device: int
+    cu_count: int
+    arch: str
+    vram_total_mb: int
+    gemm_dim: int
+    gemm_ms: float
+    gemm_tflops: float
+    ref_ms: float
+    max_abs_err: float
+    checksum: int
+    lds_ms: float
+    lds_failures: int
+    hbm_ms: float
+    hbm_gbps: float
+    peer_count: int
+    peers_accessible: int
+    ok: bool
+    error: str = ""
+
+    @classmethod
+    def from_c(cls, c: _CReport) -> "AttestReport":
+        return cls(
+            device=c.device,
+            cu_count=c.cu_count,
+            arch=c.arch.decode(errors="replace"),
+            vram_total_mb=c.vram_total_mb,
+            gemm_dim=c.gemm_m,
+            gemm_ms=c.gemm_ms,
+            gemm_tflops=c.gemm_tflops,
+            ref_ms=c.ref_ms,
+            max_abs_err=c.max_abs_err,
+            checksum=c.checksum,
+            lds_ms=c.lds_ms,
+            lds_failures=c.lds_failures,
+            hbm_ms=c.hbm_ms,
+            hbm_gbps=c.hbm_gbps,
+            peer_count=c.peer_count,
+            peers_accessible=c.peers_accessible,
+            ok=bool(c.ok),
+            error=c.error.decode(errors="replace"),
+        )
+
+
+def _load() -> ctypes.CDLL:
+    global _lib
+    if _lib is not None:
+        return _lib
+    if not _LIB_PATH.exists():
+        raise AttestationError(
+            f"attestation library missing: {_LIB_PATH} — run "
+            "python -m k8s_cc_manager_amd.ops.build"
+        )
+    lib = ctypes.CDLL(str(_LIB_PATH))
+    lib.cc_device_count.restype = ctypes.c_int
+    lib.cc_device_index_for_bdf.restype = ctypes.c_int
+    lib.cc_device_index_for_bdf.argtypes = [ctypes.c_int] * 3
+    lib.cc_attest_device.restype = ctypes.c_int
+    lib.cc_attest_device.argtypes = [
+        ctypes.c_int,
+        ctypes.c_int,
+        ctypes.POINTER(_CReport),
+    ]
+    lib.cc_mfma_gemm_bf16.restype = ctypes.c_int
+    lib.cc_mfma_gemm_bf16.argtypes = [
+        ctypes.c_int,
+        ctypes.c_void_p,
+        ctypes.c_void_p,
+        ctypes.c_void_p,
+        ctypes.c_int,
+        ctypes.c_int,
+        ctypes.c_int,
+    ]
+    lib.cc_ref_gemm_f32.restype = ctypes.c_int
+    lib.cc_ref_gemm_f32.argtypes = lib.cc_mfma_gemm_bf16.argtypes
+    _lib = lib
+    return lib
+
+
+def probe_available() -> bool:
+    """True when the library loads AND at least one GPU is visible."""
+    try:
+        return _load().cc_device_count() > 0
+    except Exception:
+        return False
+
+
+def device_count() -> int:
+    return _load().cc_device_count()
+
+
+def attest_device(device_index: int, gemm_dim: int = 1024) -> AttestReport:
+    """Run the full MFMA+LDS+HBM+xGMI probe on one GPU; raise on failure."""
+    lib = _load()
+    c = _CReport()
+    rc = lib.cc_attest_device(device_index, gemm_dim, ctypes.byref(c))
+    rep = AttestReport.from_c(c)
+    if rc != 0:
+        raise AttestationError(
+            f"device {device_index}: probe runtime error rc={rc}: {rep.error}"
+        )
+    if not rep.ok:
+        raise AttestationError(
+            f"device {device_index}: attestation FAILED "
+            f"(max_abs_err={rep.max_abs_err}, lds_failures={rep.lds_failures})"
+        )
+    logger.info(
+        "attested device %d: %s %d CUs, GEMM %.1f TF/s (%dx%dx%d, %.2f ms), "
+        "LDS ok, HBM %.0f GB/s, %d/%d xGMI peers",
+        rep.device,
+        rep.arch,
+        rep.cu_count,
+        rep.gemm_tflops,
+        rep.gemm_dim,
+        rep.gemm_dim,
+        rep.gemm_dim,
+        rep.gemm_ms,
+        rep.hbm_gbps,
+        rep.peers_accessible,
+        rep.peer_count,
+    )
+    return rep
+
+
+_BDF_RE = re.compile(
+    r"^(?:(?P<domain>[0-9a-fA-F]{4}):)?(?P<bus>[0-9a-fA-F]{2}):(?P<dev>[0-9a-fA-F]{2})\.(?P<fn>[0-7])$"
+)
+
+
+def device_index_for_bdf(bdf: str) -> int:
+    m = _BDF_RE.match(bdf.strip())
+    if not m:
+        raise AttestationError(f"unparseable bdf {bdf!r}")
+    lib = _load()
+    idx = lib.cc_device_index_for_bdf(
+        int(m.group("domain") or "0", 16),
+        int(m.group("bus"), 16),
+        int(m.group("dev"), 16),
+    )
+    if idx < 0:
+        raise AttestationError(f"no HIP device with bdf {bdf}")
+    return idx
+
+
+def attest_device_by_bdf(device) -> None:
+    """TransitionEngine attestor hook: CCDevice -> None (raises on fail).
+
+    The probe gates cc.ready.state: a GPU that resets but cannot run
+    MFMA/LDS work correctly must not be labeled ready
+    (replaces the register-readback-only verify of the reference,
+    /root/reference/main.py:523-529).
+    """
+    gemm_dim = int(os.environ.get("CC_ATTEST_GEMM_DIM", "1024"))
+    idx = device_index_for_bdf(device.bdf)
+    attest_device(idx, gemm_dim=gemm_dim)
+
+
+def mfma_gemm_bf16(device_index: int, a_ptr: int, bt_ptr: int, c_ptr: int,
+                   m: int, n: int, k: int) -> None:
+    """Launch C[M,N] = A[M,K] @ Bt[N,K]^T on caller-owned device buffers
+    (torch tensors via .data_ptr()). M,N multiples of 128, K of 32."""
+    rc = _load().cc_mfma_gemm_bf16(device_index, a_ptr, bt_ptr, c_ptr, m, n, k)
+    if rc != 0:
+        raise AttestationError(f"mfma_gemm_bf16 rc={rc}")
+
+
+def ref_gemm_f32(device_index: int, a_ptr: int, bt_ptr: int, c_ptr: int,
+                 m: int, n: int, k: int) -> None:
+    rc = _load().cc_ref_gemm_f32(device_index, a_ptr, bt_ptr, c_ptr, m, n, k)
+    if rc != 0:
+        raise AttestationError(f"ref_gemm_f32 rc={rc}")

@@ -1,0 +1,480 @@
+// CDNA4 (gfx950 / MI355X) post-reset attestation probe.
+//
+// The one hand-written kernel family of the AMD CC manager. After a CC
+// transition resets a GPU, the reference merely re-reads the mode
+// register (/root/reference/main.py:523-529). Here the device must
+// additionally PROVE it executes correctly inside the TEE:
+//
+//   1. mfma_gemm_bf16   — LDS-tiled bf16 GEMM on the MFMA matrix cores
+//                         (v_mfma_f32_16x16x32_bf16), exercising HBM ->
+//                         LDS -> VGPR -> MFMA -> HBM end to end;
+//   2. ref_gemm_f32     — plain VALU fp32 GEMM of the same inputs: the
+//                         independent on-device ground truth (inputs are
+//                         small integers, so both paths must agree
+//                         BITWISE — any mismatch is silicon/TEE trouble);
+//   3. lds_probe        — LDS cell sweep with rotating patterns;
+//   4. hbm_probe        — vectorized streaming copy + checksum (HBM3E
+//                         path, reported as GB/s);
+//   5. xGMI peer visibility via hipDeviceCanAccessPeer.
+//
+// Layout notes (from the CDNA4 guides): wave = 64 lanes; for
+// v_mfma_f32_16x16x32_bf16 each lane carries 8 bf16 of A and B and 4
+// fp32 of C/D; A lane mapping row=l&15, k=(l>>4)*8+j; B (we require the
+// second operand TRANSPOSED, i.e. Bt[N][K], so its lane mapping is the
+// same as A's with col=l&15); C/D mapping col=l&15, row=(l>>4)*4+r.
+// LDS tiles are padded to a 36-element (72 B) row stride so the 16-lane
+// groups of ds_read_b128 land on 16 distinct banks ((18*r + c) mod 64
+// is injective over r=0..15).
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#include <cstdint>
+#include <cstdio>
+
+#define CC_CHECK(expr)                                                         \
+  do {                                                                         \
+    hipError_t _e = (expr);                                                    \
+    if (_e != hipSuccess) {                                                    \
+      snprintf(rep->error, sizeof(rep->error), "%s: %s", #expr,                \
+               hipGetErrorString(_e));                                         \
+      return (int)_e;                                                          \
+    }                                                                          \
+  } while (0)
+
+using bf16 = __bf16;
+typedef __attribute__((ext_vector_type(8))) bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(4))) float float4v;
+
+// ---------------------------------------------------------------------------
+// Tiling parameters: 128x128 block tile, 4 waves in a 2x2 grid, each wave
+// owns a 64x64 sub-tile = 4x4 MFMA tiles of 16x16, K staged in 32-deep
+// chunks through LDS (double-buffered).
+// ---------------------------------------------------------------------------
+constexpr int BM = 128;       // block tile rows
+constexpr int BN = 128;       // block tile cols
+constexpr int BK = 32;        // K chunk per LDS stage
+constexpr int LDS_STRIDE = 36;  // bf16 elements per padded LDS row (32 + 4)
+
+// ---------------------------------------------------------------------------
+// deterministic small-integer fill: values in {-2,-1,0,1}; products and
+// K<=8192 partial sums stay exact in fp32 AND in the MFMA accumulator,
+// so the MFMA and VALU paths must agree bitwise.
+// ---------------------------------------------------------------------------
+__global__ void fill_bf16_lcg(bf16* __restrict__ out, long n, uint32_t seed) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    uint32_t x = (uint32_t)i * 1103515245u + seed * 747796405u + 12345u;
+    x ^= x >> 16;
+    x *= 2654435769u;
+    int v = (int)((x >> 13) & 3u) - 2;  // {-2,-1,0,1}
+    out[i] = (bf16)(float)v;
+  }
+}
+
+__global__ void fill_f32_iota(float* __restrict__ out, long n, float scale) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) out[i] = scale * (float)(i & 1023);
+}
+
+// ---------------------------------------------------------------------------
+// MFMA bf16 GEMM:  C[M,N] = A[M,K] @ Bt[N,K]^T      (all row-major)
+// Grid: (N/BN, M/BM); block: 256 threads (4 waves, 2x2 of 64x64).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256, 2) void mfma_gemm_bf16(
+    const bf16* __restrict__ A, const bf16* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K) {
+  __shared__ bf16 lds[2][2][BM * LDS_STRIDE];  // [buf][A/B][tile]
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;          // 0..3
+  const int wave_m = (wave >> 1) * 64;  // wave row offset in block tile
+  const int wave_n = (wave & 1) * 64;   // wave col offset
+  const int block_m = blockIdx.y * BM;
+  const int block_n = blockIdx.x * BN;
+
+  // Staging: 256 threads, tile is 128 rows x 32 bf16 (64 B) per operand.
+  // Each thread copies 16 B: 256 * 16 B = 4 KB per pass; tile is 8 KB ->
+  // 2 passes per operand. Thread t of pass p loads row r = t/4 (+64*p),
+  // 8 consecutive bf16 at col (t%4)*8.
+  const int ld_row = tid >> 2;          // 0..63
+  const int ld_col = (tid & 3) * 8;     // 0,8,16,24
+
+  f32x4 acc[4][4] = {};  // 4x4 MFMA tiles of 16x16 per wave
+
+  const int a_row = lane & 15;          // fragment row within 16-row tile
+  const int a_koff = (lane >> 4) * 8;   // fragment k offset (0,8,16,24)
+
+  int nk = K / BK;
+  for (int kt = 0; kt < nk; ++kt) {
+    const int buf = kt & 1;
+    const int k0 = kt * BK;
+    // ---- stage A and Bt tiles into LDS -------------------------------
+    bf16* As = lds[buf][0];
+    bf16* Bs = lds[buf][1];
+#pragma unroll
+    for (int p = 0; p < 2; ++p) {
+      int r = ld_row + 64 * p;
+      // global: A[block_m + r][k0 + ld_col .. +8]
+      const bf16* ga = A + (long)(block_m + r) * K + k0 + ld_col;
+      const bf16* gb = Bt + (long)(block_n + r) * K + k0 + ld_col;
+      *(bf16x8*)&As[r * LDS_STRIDE + ld_col] = *(const bf16x8*)ga;
+      *(bf16x8*)&Bs[r * LDS_STRIDE + ld_col] = *(const bf16x8*)gb;
+    }
+    __syncthreads();
+
+    // ---- MFMA over the 32-deep chunk: one 16x16x32 issue per tile
+    // pair (the lane fragment k = (lane>>4)*8 + j spans k=0..31 across
+    // the four 16-lane groups, exactly one BK chunk).
+    {
+      bf16x8 afrag[4], bfrag[4];
+#pragma unroll
+      for (int t = 0; t < 4; ++t) {
+        int ar = wave_m + t * 16 + a_row;
+        int bc = wave_n + t * 16 + a_row;  // same in-tile mapping for Bt
+        afrag[t] = *(const bf16x8*)&As[ar * LDS_STRIDE + a_koff];
+        bfrag[t] = *(const bf16x8*)&Bs[bc * LDS_STRIDE + a_koff];
+      }
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: C/D mapping col=lane&15, row=(lane>>4)*4+r ----------
+  const int c_col = lane & 15;
+  const int c_row0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = block_m + wave_m + i * 16 + c_row0 + r;
+        int col = block_n + wave_n + j * 16 + c_col;
+        C[(long)row * N + col] = acc[i][j][r];
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// VALU fp32 reference GEMM (independent ground truth; deliberately does
+// NOT share tiling or fragment code with the MFMA path).
+// One thread per C element, fp32 FMA chain over K.
+// ---------------------------------------------------------------------------
+__global__ void ref_gemm_f32(const bf16* __restrict__ A,
+                             const bf16* __restrict__ Bt,
+                             float* __restrict__ C, int M, int N, int K) {
+  int col = blockIdx.x * blockDim.x + threadIdx.x;
+  int row = blockIdx.y * blockDim.y + threadIdx.y;
+  if (row >= M || col >= N) return;
+  float acc = 0.f;
+  const bf16* a = A + (long)row * K;
+  const bf16* b = Bt + (long)col * K;
+  for (int k = 0; k < K; ++k) acc = fmaf((float)a[k], (float)b[k], acc);
+  C[(long)row * N + col] = acc;
+}
+
+// max |x-y| reduction over n elements -> out[0] (pre-zeroed)
+__global__ void max_abs_diff(const float* __restrict__ x,
+                             const float* __restrict__ y, long n,
+                             float* out) {
+  __shared__ float smax[256];
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  float m = 0.f;
+  for (; i < n; i += stride) {
+    float d = fabsf(x[i] - y[i]);
+    m = fmaxf(m, d);
+  }
+  smax[threadIdx.x] = m;
+  __syncthreads();
+  for (int s = 128; s > 0; s >>= 1) {
+    if (threadIdx.x < s) smax[threadIdx.x] = fmaxf(smax[threadIdx.x], smax[threadIdx.x + s]);
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    // fp32 max via atomicMax on the bit pattern (values are >= 0)
+    atomicMax((unsigned int*)out, __float_as_uint(smax[0]));
+  }
+}
+
+// FNV-1a style checksum of a float buffer -> 64-bit xor-fold (order
+// independent via per-element mix, so concurrent blocks are fine).
+__global__ void checksum_f32(const float* __restrict__ x, long n,
+                             unsigned long long* out) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  unsigned long long h = 0;
+  for (; i < n; i += stride) {
+    unsigned long long v = (unsigned long long)__float_as_uint(x[i]) + 0x9e3779b97f4a7c15ull * (unsigned long long)(i + 1);
+    v ^= v >> 33; v *= 0xff51afd7ed558ccdull; v ^= v >> 33;
+    h ^= v;
+  }
+  // xor-reduce across the wave then one atomic per wave
+  for (int off = 32; off > 0; off >>= 1)
+    h ^= __shfl_down(h, off, 64);
+  if ((threadIdx.x & 63) == 0) atomicXor(out, h);
+}
+
+// ---------------------------------------------------------------------------
+// LDS probe: rotating write/read patterns across a 64 KiB LDS slab per
+// workgroup; validates every cell and measures aggregate LDS traffic.
+// ---------------------------------------------------------------------------
+constexpr int LDS_PROBE_WORDS = 16384;  // 64 KiB of uint32 per workgroup
+
+__global__ void lds_probe(uint32_t* __restrict__ fail_count, int rounds) {
+  __shared__ uint32_t slab[LDS_PROBE_WORDS];
+  const int tid = threadIdx.x;
+  const int nthreads = blockDim.x;
+  uint32_t local_fail = 0;
+  for (int r = 0; r < rounds; ++r) {
+    uint32_t salt = 0x9e3779b9u * (r + 1) + blockIdx.x;
+    for (int i = tid; i < LDS_PROBE_WORDS; i += nthreads)
+      slab[i] = (uint32_t)i * 2654435769u + salt;
+    __syncthreads();
+    // read back with a different (conflict-heavy, bank-rotating) stride
+    for (int i = tid; i < LDS_PROBE_WORDS; i += nthreads) {
+      int j = (i * 33 + r) & (LDS_PROBE_WORDS - 1);
+      uint32_t want = (uint32_t)j * 2654435769u + salt;
+      if (slab[j] != want) ++local_fail;
+    }
+    __syncthreads();
+  }
+  if (local_fail) atomicAdd(fail_count, local_fail);
+}
+
+// ---------------------------------------------------------------------------
+// HBM probe: float4 streaming copy (the measured-best pattern, ~79% of
+// the 8 TB/s peak per MI355X_MICROARCH.md) + spot validation.
+// ---------------------------------------------------------------------------
+__global__ void hbm_copy_f4(const float4v* __restrict__ src,
+                            float4v* __restrict__ dst, long n4) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (; i < n4; i += stride) dst[i] = src[i];
+}
+
+// ===========================================================================
+// C API
+// ===========================================================================
+extern "C" {
+
+struct CcAttestReport {
+  int device;
+  int cu_count;
+  int xcc_count;
+  char arch[64];
+  char error[256];
+  long long vram_total_mb;
+  // GEMM probe
+  int gemm_m, gemm_n, gemm_k;
+  double gemm_ms;
+  double gemm_tflops;
+  double ref_ms;
+  float max_abs_err;       // MFMA vs VALU fp32 (must be 0.0: exact inputs)
+  unsigned long long checksum;
+  // LDS probe
+  double lds_ms;
+  unsigned int lds_failures;
+  // HBM probe
+  double hbm_ms;
+  double hbm_gbps;
+  // fabric
+  int peer_count;          // devices visible
+  int peers_accessible;    // peers with canAccessPeer==1
+  int ok;
+};
+
+// Minimal liveness check: a real kernel launch must complete. Used as
+// the post-reset boot-wait gate (a device can answer property queries
+// from cached driver state while its command processor is wedged).
+__global__ void liveness_kernel(int* out) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) *out = 0x600D;
+}
+
+int cc_device_alive(int device) {
+  if (hipSetDevice(device) != hipSuccess) return -1;
+  int* d = nullptr;
+  if (hipMalloc(&d, sizeof(int)) != hipSuccess) return -2;
+  hipLaunchKernelGGL(liveness_kernel, dim3(1), dim3(64), 0, 0, d);
+  int h = 0;
+  hipError_t e = hipMemcpy(&h, d, sizeof(int), hipMemcpyDeviceToHost);
+  (void)hipFree(d);
+  if (e != hipSuccess) return -3;
+  return h == 0x600D ? 0 : -4;
+}
+
+int cc_device_count(void) {
+  int n = 0;
+  if (hipGetDeviceCount(&n) != hipSuccess) return -1;
+  return n;
+}
+
+int cc_device_index_for_bdf(int domain, int bus, int dev) {
+  int n = 0;
+  if (hipGetDeviceCount(&n) != hipSuccess) return -1;
+  for (int i = 0; i < n; ++i) {
+    hipDeviceProp_t p;
+    if (hipGetDeviceProperties(&p, i) != hipSuccess) continue;
+    if (p.pciDomainID == domain && p.pciBusID == bus && p.pciDeviceID == dev)
+      return i;
+  }
+  return -1;
+}
+
+// Run the bf16 MFMA GEMM on caller-provided device buffers (e.g. torch
+// tensors): C[M,N] = A[M,K] @ Bt[N,K]^T. M,N multiples of 128, K of 32.
+int cc_mfma_gemm_bf16(int device, const void* A, const void* Bt, void* C,
+                      int M, int N, int K) {
+  if (M % BM || N % BN || K % BK) return -2;
+  if (hipSetDevice(device) != hipSuccess) return -3;
+  dim3 grid(N / BN, M / BM);
+  hipLaunchKernelGGL(mfma_gemm_bf16, grid, dim3(256), 0, 0, (const bf16*)A,
+                     (const bf16*)Bt, (float*)C, M, N, K);
+  return (int)hipDeviceSynchronize();
+}
+
+// Plain fp32 reference GEMM on the same operand convention.
+int cc_ref_gemm_f32(int device, const void* A, const void* Bt, void* C,
+                    int M, int N, int K) {
+  if (hipSetDevice(device) != hipSuccess) return -3;
+  dim3 block(16, 16);
+  dim3 grid((N + 15) / 16, (M + 15) / 16);
+  hipLaunchKernelGGL(ref_gemm_f32, grid, block, 0, 0, (const bf16*)A,
+                     (const bf16*)Bt, (float*)C, M, N, K);
+  return (int)hipDeviceSynchronize();
+}
+
+static double event_ms(hipEvent_t a, hipEvent_t b) {
+  float ms = 0.f;
+  hipEventElapsedTime(&ms, a, b);
+  return (double)ms;
+}
+
+// The full self-contained attestation probe. gemm_dim: problem size
+// (square); 1024 for the post-reset gate (fast), 4096+ for perf
+// characterization. rep->ok = 1 iff every stage validated.
+int cc_attest_device(int device, int gemm_dim, struct CcAttestReport* rep) {
+  if (!rep) return -1;
+  __builtin_memset(rep, 0, sizeof(*rep));
+  rep->device = device;
+  CC_CHECK(hipSetDevice(device));
+
+  hipDeviceProp_t prop;
+  CC_CHECK(hipGetDeviceProperties(&prop, device));
+  rep->cu_count = prop.multiProcessorCount;
+  rep->vram_total_mb = (long long)(prop.totalGlobalMem >> 20);
+  __builtin_strncpy(rep->arch, prop.gcnArchName, sizeof(rep->arch) - 1);
+
+  const int D = (gemm_dim / BM) * BM > 0 ? (gemm_dim / BM) * BM : BM;
+  rep->gemm_m = rep->gemm_n = rep->gemm_k = D;
+  long elems = (long)D * D;
+
+  bf16 *dA = nullptr, *dB = nullptr;
+  float *dC = nullptr, *dRef = nullptr, *dErr = nullptr;
+  unsigned long long* dSum = nullptr;
+  CC_CHECK(hipMalloc(&dA, elems * sizeof(bf16)));
+  CC_CHECK(hipMalloc(&dB, elems * sizeof(bf16)));
+  CC_CHECK(hipMalloc(&dC, elems * sizeof(float)));
+  CC_CHECK(hipMalloc(&dRef, elems * sizeof(float)));
+  CC_CHECK(hipMalloc(&dErr, sizeof(float)));
+  CC_CHECK(hipMalloc(&dSum, sizeof(unsigned long long)));
+  CC_CHECK(hipMemset(dErr, 0, sizeof(float)));
+  CC_CHECK(hipMemset(dSum, 0, sizeof(unsigned long long)));
+
+  hipEvent_t ev0, ev1;
+  CC_CHECK(hipEventCreate(&ev0));
+  CC_CHECK(hipEventCreate(&ev1));
+
+  // -- fill (asymmetric seeds: catches row/col-swapped layouts) --------
+  hipLaunchKernelGGL(fill_bf16_lcg, dim3(2048), dim3(256), 0, 0, dA, elems, 1u);
+  hipLaunchKernelGGL(fill_bf16_lcg, dim3(2048), dim3(256), 0, 0, dB, elems, 7u);
+
+  // -- MFMA GEMM (timed; warm once) ------------------------------------
+  dim3 grid(D / BN, D / BM);
+  hipLaunchKernelGGL(mfma_gemm_bf16, grid, dim3(256), 0, 0, dA, dB, dC, D, D, D);
+  CC_CHECK(hipDeviceSynchronize());
+  CC_CHECK(hipEventRecord(ev0, 0));
+  hipLaunchKernelGGL(mfma_gemm_bf16, grid, dim3(256), 0, 0, dA, dB, dC, D, D, D);
+  CC_CHECK(hipEventRecord(ev1, 0));
+  CC_CHECK(hipEventSynchronize(ev1));
+  rep->gemm_ms = event_ms(ev0, ev1);
+  rep->gemm_tflops = 2.0 * D * (double)D * D / (rep->gemm_ms * 1e-3) / 1e12;
+
+  // -- VALU reference + compare ---------------------------------------
+  CC_CHECK(hipEventRecord(ev0, 0));
+  {
+    dim3 rblock(16, 16), rgrid((D + 15) / 16, (D + 15) / 16);
+    hipLaunchKernelGGL(ref_gemm_f32, rgrid, rblock, 0, 0, dA, dB, dRef, D, D, D);
+  }
+  CC_CHECK(hipEventRecord(ev1, 0));
+  CC_CHECK(hipEventSynchronize(ev1));
+  rep->ref_ms = event_ms(ev0, ev1);
+
+  hipLaunchKernelGGL(max_abs_diff, dim3(1024), dim3(256), 0, 0, dC, dRef,
+                     elems, dErr);
+  hipLaunchKernelGGL(checksum_f32, dim3(1024), dim3(256), 0, 0, dC, elems, dSum);
+  CC_CHECK(hipDeviceSynchronize());
+  unsigned int err_bits = 0;
+  CC_CHECK(hipMemcpy(&err_bits, dErr, sizeof(err_bits), hipMemcpyDeviceToHost));
+  float max_err;
+  __builtin_memcpy(&max_err, &err_bits, sizeof(max_err));
+  rep->max_abs_err = max_err;
+  CC_CHECK(hipMemcpy(&rep->checksum, dSum, sizeof(rep->checksum),
+                     hipMemcpyDeviceToHost));
+
+  // -- LDS probe -------------------------------------------------------
+  uint32_t* dFail = nullptr;
+  CC_CHECK(hipMalloc(&dFail, sizeof(uint32_t)));
+  CC_CHECK(hipMemset(dFail, 0, sizeof(uint32_t)));
+  CC_CHECK(hipEventRecord(ev0, 0));
+  hipLaunchKernelGGL(lds_probe, dim3(512), dim3(256), 0, 0, dFail, 8);
+  CC_CHECK(hipEventRecord(ev1, 0));
+  CC_CHECK(hipEventSynchronize(ev1));
+  rep->lds_ms = event_ms(ev0, ev1);
+  CC_CHECK(hipMemcpy(&rep->lds_failures, dFail, sizeof(uint32_t),
+                     hipMemcpyDeviceToHost));
+  hipFree(dFail);
+
+  // -- HBM probe: reuse C/Ref buffers as src/dst -----------------------
+  long n4 = elems / 4;
+  CC_CHECK(hipEventRecord(ev0, 0));
+  hipLaunchKernelGGL(hbm_copy_f4, dim3(4096), dim3(256), 0, 0,
+                     (const float4v*)dC, (float4v*)dRef, n4);
+  CC_CHECK(hipEventRecord(ev1, 0));
+  CC_CHECK(hipEventSynchronize(ev1));
+  rep->hbm_ms = event_ms(ev0, ev1);
+  rep->hbm_gbps = 2.0 * n4 * 16.0 / (rep->hbm_ms * 1e-3) / 1e9;
+
+  // -- xGMI peer visibility -------------------------------------------
+  int ndev = 0;
+  CC_CHECK(hipGetDeviceCount(&ndev));
+  rep->peer_count = ndev - 1;
+  for (int p = 0; p < ndev; ++p) {
+    if (p == device) continue;
+    int can = 0;
+    if (hipDeviceCanAccessPeer(&can, device, p) == hipSuccess && can)
+      ++rep->peers_accessible;
+  }
+
+  hipEventDestroy(ev0);
+  hipEventDestroy(ev1);
+  hipFree(dA); hipFree(dB); hipFree(dC); hipFree(dRef);
+  hipFree(dErr); hipFree(dSum);
+
+  rep->ok = (rep->max_abs_err == 0.0f) && (rep->lds_failures == 0) &&
+            (rep->gemm_tflops > 0.0) ? 1 : 0;
+  return 0;
+}
+
+}  // extern "C"
