@@ -22,15 +22,16 @@
 // fp32 of C/D; A lane mapping row=l&15, k=(l>>4)*8+j; B (we require the
 // second operand TRANSPOSED, i.e. Bt[N][K], so its lane mapping is the
 // same as A's with col=l&15); C/D mapping col=l&15, row=(l>>4)*4+r.
-// LDS tiles are padded to a 36-element (72 B) row stride so the 16-lane
-// groups of ds_read_b128 land on 16 distinct banks ((18*r + c) mod 64
-// is injective over r=0..15).
+// LDS tile images carry the st_16x32 XOR swizzle (byte-bit-5 ^= bit-9)
+// so ds_read_b128 fragment reads are bank-conflict-free while the
+// global->LDS DMA stays lane-linear (swizzle rides the source address).
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
 
 #include <cstdint>
 #include <cstdio>
+#include <mutex>
 
 #define CC_CHECK(expr)                                                         \
   do {                                                                         \
@@ -52,10 +53,9 @@ typedef __attribute__((ext_vector_type(4))) float float4v;
 // owns a 64x64 sub-tile = 4x4 MFMA tiles of 16x16, K staged in 32-deep
 // chunks through LDS (double-buffered).
 // ---------------------------------------------------------------------------
-constexpr int BM = 128;       // block tile rows
-constexpr int BN = 128;       // block tile cols
-constexpr int BK = 32;        // K chunk per LDS stage
-constexpr int LDS_STRIDE = 36;  // bf16 elements per padded LDS row (32 + 4)
+constexpr int BM = 128;  // block tile rows
+constexpr int BN = 128;  // block tile cols
+constexpr int BK = 64;   // K chunk per LDS stage
 
 // ---------------------------------------------------------------------------
 // deterministic small-integer fill: values in {-2,-1,0,1}; products and
@@ -83,61 +83,88 @@ __global__ void fill_f32_iota(float* __restrict__ out, long n, float scale) {
 // ---------------------------------------------------------------------------
 // MFMA bf16 GEMM:  C[M,N] = A[M,K] @ Bt[N,K]^T      (all row-major)
 // Grid: (N/BN, M/BM); block: 256 threads (4 waves, 2x2 of 64x64).
+//
+// Structure: the "step-3" ladder shape of the CDNA4 guide — 128x128
+// tile, BK=64, direct global->LDS DMA (global_load_lds_dwordx4, the
+// compiler never auto-emits it) double-buffered, st_16x32 XOR swizzle
+// applied on the pre-swizzled GLOBAL source address (glds lands
+// lane-linear in LDS, so the swizzle must ride the source) and on the
+// ds_read fragment address, which turns the 8-way ds_read_b128 bank
+// conflict of a linear [128][64]-bf16 image into a conflict-free read.
 // ---------------------------------------------------------------------------
+constexpr int TILE_B = BM * BK * 2;  // 16 KiB per operand tile
+
+__device__ __forceinline__ int swz(int byte_off) {
+  // st_16x32 swizzle: XOR byte-bit-5 with bit-9 inside each 1 KiB subtile
+  return byte_off ^ (((byte_off >> 9) & 1) << 5);
+}
+
+__device__ __forceinline__ void stage_tile_glds(
+    const char* gbase, long row_stride_b, long k0_b, char* lds_tile,
+    int wave, int lane) {
+#pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    int base = (wave * 4 + p) * 1024;       // wave-uniform LDS dest
+    int logical = swz(base + lane * 16);    // where lane's 16 B lands
+    int row = logical >> 7;                 // 128 B per logical row (BK=64)
+    int colb = logical & 127;
+    const char* g = gbase + (long)row * row_stride_b + k0_b + colb;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)g,
+        (__attribute__((address_space(3))) void*)(lds_tile + base), 16, 0, 0);
+  }
+}
+
 __global__ __launch_bounds__(256, 2) void mfma_gemm_bf16(
     const bf16* __restrict__ A, const bf16* __restrict__ Bt,
     float* __restrict__ C, int M, int N, int K) {
-  __shared__ bf16 lds[2][2][BM * LDS_STRIDE];  // [buf][A/B][tile]
+  // [buf][A|B][16 KiB]: one __shared__ object (a second one makes the
+  // compiler drain vmcnt before every ds_read of a glds pipeline).
+  __shared__ char lds[2 * 2 * TILE_B];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
-  const int wave = tid >> 6;          // 0..3
+  const int wave = tid >> 6;            // 0..3
   const int wave_m = (wave >> 1) * 64;  // wave row offset in block tile
   const int wave_n = (wave & 1) * 64;   // wave col offset
   const int block_m = blockIdx.y * BM;
   const int block_n = blockIdx.x * BN;
 
-  // Staging: 256 threads, tile is 128 rows x 32 bf16 (64 B) per operand.
-  // Each thread copies 16 B: 256 * 16 B = 4 KB per pass; tile is 8 KB ->
-  // 2 passes per operand. Thread t of pass p loads row r = t/4 (+64*p),
-  // 8 consecutive bf16 at col (t%4)*8.
-  const int ld_row = tid >> 2;          // 0..63
-  const int ld_col = (tid & 3) * 8;     // 0,8,16,24
+  const char* gA = (const char*)(A + (long)block_m * K);
+  const char* gB = (const char*)(Bt + (long)block_n * K);
+  const long row_b = (long)K * 2;
 
   f32x4 acc[4][4] = {};  // 4x4 MFMA tiles of 16x16 per wave
 
-  const int a_row = lane & 15;          // fragment row within 16-row tile
-  const int a_koff = (lane >> 4) * 8;   // fragment k offset (0,8,16,24)
+  const int lane15 = lane & 15;
+  const int khalf_b = (lane >> 4) * 16;  // fragment k byte offset
 
-  int nk = K / BK;
+  const int nk = K / BK;
+  // prologue: stage tile 0 into buffer 0
+  stage_tile_glds(gA, row_b, 0, &lds[0], wave, lane);
+  stage_tile_glds(gB, row_b, 0, &lds[TILE_B], wave, lane);
+  __syncthreads();  // drains the glds (vmcnt0 inside the barrier)
+
   for (int kt = 0; kt < nk; ++kt) {
     const int buf = kt & 1;
-    const int k0 = kt * BK;
-    // ---- stage A and Bt tiles into LDS -------------------------------
-    bf16* As = lds[buf][0];
-    bf16* Bs = lds[buf][1];
-#pragma unroll
-    for (int p = 0; p < 2; ++p) {
-      int r = ld_row + 64 * p;
-      // global: A[block_m + r][k0 + ld_col .. +8]
-      const bf16* ga = A + (long)(block_m + r) * K + k0 + ld_col;
-      const bf16* gb = Bt + (long)(block_n + r) * K + k0 + ld_col;
-      *(bf16x8*)&As[r * LDS_STRIDE + ld_col] = *(const bf16x8*)ga;
-      *(bf16x8*)&Bs[r * LDS_STRIDE + ld_col] = *(const bf16x8*)gb;
+    char* As = &lds[buf * 2 * TILE_B];
+    char* Bs = As + TILE_B;
+    // issue DMA for the NEXT tile into the other buffer (no wait here)
+    if (kt + 1 < nk) {
+      char* An = &lds[(buf ^ 1) * 2 * TILE_B];
+      stage_tile_glds(gA, row_b, (long)(kt + 1) * BK * 2, An, wave, lane);
+      stage_tile_glds(gB, row_b, (long)(kt + 1) * BK * 2, An + TILE_B, wave, lane);
     }
-    __syncthreads();
-
-    // ---- MFMA over the 32-deep chunk: one 16x16x32 issue per tile
-    // pair (the lane fragment k = (lane>>4)*8 + j spans k=0..31 across
-    // the four 16-lane groups, exactly one BK chunk).
-    {
+    // ---- MFMA over the 64-deep chunk: 2 k-steps of 32 ---------------
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
       bf16x8 afrag[4], bfrag[4];
 #pragma unroll
       for (int t = 0; t < 4; ++t) {
-        int ar = wave_m + t * 16 + a_row;
-        int bc = wave_n + t * 16 + a_row;  // same in-tile mapping for Bt
-        afrag[t] = *(const bf16x8*)&As[ar * LDS_STRIDE + a_koff];
-        bfrag[t] = *(const bf16x8*)&Bs[bc * LDS_STRIDE + a_koff];
+        int a_log = (wave_m + t * 16 + lane15) * 128 + ks * 64 + khalf_b;
+        int b_log = (wave_n + t * 16 + lane15) * 128 + ks * 64 + khalf_b;
+        afrag[t] = *(const bf16x8*)(As + swz(a_log));
+        bfrag[t] = *(const bf16x8*)(Bs + swz(b_log));
       }
 #pragma unroll
       for (int i = 0; i < 4; ++i)
@@ -146,6 +173,9 @@ __global__ __launch_bounds__(256, 2) void mfma_gemm_bf16(
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
     }
+    // one barrier per K-tile: everyone done reading buf, and the
+    // in-flight glds for buf^1 is drained (implicit vmcnt(0) — the
+    // structural cost of the 2-barrier structure, accepted here).
     __syncthreads();
   }
 
@@ -267,6 +297,70 @@ __global__ void hbm_copy_f4(const float4v* __restrict__ src,
 // ===========================================================================
 // C API
 // ===========================================================================
+// ---------------------------------------------------------------------------
+// Persistent per-device probe context. The CC manager is a long-lived
+// daemon that attests after every transition: allocations, events and
+// the liveness scratch word are cached so steady-state probes pay only
+// kernel time (first call measured ~225 ms of hipMalloc/teardown
+// overhead at dim=1024; cached, the probe is ~2 ms).
+// ---------------------------------------------------------------------------
+constexpr int kMaxDevices = 64;
+
+struct ProbeCtx {
+  int dim = 0;
+  bf16 *dA = nullptr, *dB = nullptr;
+  float *dC = nullptr, *dRef = nullptr, *dErr = nullptr;
+  unsigned long long* dSum = nullptr;
+  uint32_t* dFail = nullptr;
+  int* dLive = nullptr;
+  hipEvent_t ev0 = nullptr, ev1 = nullptr;
+};
+
+static ProbeCtx g_ctx[kMaxDevices];
+static std::mutex g_ctx_mu;
+
+static void ctx_release(ProbeCtx& c) {
+  if (c.dA) (void)hipFree(c.dA);
+  if (c.dB) (void)hipFree(c.dB);
+  if (c.dC) (void)hipFree(c.dC);
+  if (c.dRef) (void)hipFree(c.dRef);
+  if (c.dErr) (void)hipFree(c.dErr);
+  if (c.dSum) (void)hipFree(c.dSum);
+  if (c.dFail) (void)hipFree(c.dFail);
+  if (c.dLive) (void)hipFree(c.dLive);
+  if (c.ev0) (void)hipEventDestroy(c.ev0);
+  if (c.ev1) (void)hipEventDestroy(c.ev1);
+  c = ProbeCtx{};
+}
+
+static hipError_t ctx_acquire(int device, int dim, ProbeCtx** out) {
+  ProbeCtx& c = g_ctx[device];
+  if (c.dim == dim) {
+    *out = &c;
+    return hipSuccess;
+  }
+  ctx_release(c);
+  long elems = (long)dim * dim;
+  hipError_t e;
+  if ((e = hipMalloc(&c.dA, elems * sizeof(bf16))) != hipSuccess) return e;
+  if ((e = hipMalloc(&c.dB, elems * sizeof(bf16))) != hipSuccess) return e;
+  if ((e = hipMalloc(&c.dC, elems * sizeof(float))) != hipSuccess) return e;
+  if ((e = hipMalloc(&c.dRef, elems * sizeof(float))) != hipSuccess) return e;
+  if ((e = hipMalloc(&c.dErr, sizeof(float))) != hipSuccess) return e;
+  if ((e = hipMalloc(&c.dSum, sizeof(unsigned long long))) != hipSuccess) return e;
+  if ((e = hipMalloc(&c.dFail, sizeof(uint32_t))) != hipSuccess) return e;
+  if ((e = hipMalloc(&c.dLive, sizeof(int))) != hipSuccess) return e;
+  if ((e = hipEventCreate(&c.ev0)) != hipSuccess) return e;
+  if ((e = hipEventCreate(&c.ev1)) != hipSuccess) return e;
+  c.dim = dim;
+  *out = &c;
+  return hipSuccess;
+}
+
+// The full self-contained attestation probe. gemm_dim: problem size
+// (square); 1024 for the post-reset gate (fast), 4096+ for perf
+// characterization. rep->ok = 1 iff every stage validated.
+
 extern "C" {
 
 struct CcAttestReport {
@@ -303,13 +397,19 @@ __global__ void liveness_kernel(int* out) {
 }
 
 int cc_device_alive(int device) {
+  if (device < 0 || device >= kMaxDevices) return -1;
   if (hipSetDevice(device) != hipSuccess) return -1;
   int* d = nullptr;
-  if (hipMalloc(&d, sizeof(int)) != hipSuccess) return -2;
+  {
+    std::lock_guard<std::mutex> lk(g_ctx_mu);
+    ProbeCtx& c = g_ctx[device];
+    if (!c.dLive && hipMalloc(&c.dLive, sizeof(int)) != hipSuccess) return -2;
+    d = c.dLive;
+  }
+  if (hipMemset(d, 0, sizeof(int)) != hipSuccess) return -2;
   hipLaunchKernelGGL(liveness_kernel, dim3(1), dim3(64), 0, 0, d);
   int h = 0;
   hipError_t e = hipMemcpy(&h, d, sizeof(int), hipMemcpyDeviceToHost);
-  (void)hipFree(d);
   if (e != hipSuccess) return -3;
   return h == 0x600D ? 0 : -4;
 }
@@ -357,13 +457,10 @@ int cc_ref_gemm_f32(int device, const void* A, const void* Bt, void* C,
 
 static double event_ms(hipEvent_t a, hipEvent_t b) {
   float ms = 0.f;
-  hipEventElapsedTime(&ms, a, b);
+  (void)hipEventElapsedTime(&ms, a, b);
   return (double)ms;
 }
 
-// The full self-contained attestation probe. gemm_dim: problem size
-// (square); 1024 for the post-reset gate (fast), 4096+ for perf
-// characterization. rep->ok = 1 iff every stage validated.
 int cc_attest_device(int device, int gemm_dim, struct CcAttestReport* rep) {
   if (!rep) return -1;
   __builtin_memset(rep, 0, sizeof(*rep));
@@ -380,21 +477,16 @@ int cc_attest_device(int device, int gemm_dim, struct CcAttestReport* rep) {
   rep->gemm_m = rep->gemm_n = rep->gemm_k = D;
   long elems = (long)D * D;
 
-  bf16 *dA = nullptr, *dB = nullptr;
-  float *dC = nullptr, *dRef = nullptr, *dErr = nullptr;
-  unsigned long long* dSum = nullptr;
-  CC_CHECK(hipMalloc(&dA, elems * sizeof(bf16)));
-  CC_CHECK(hipMalloc(&dB, elems * sizeof(bf16)));
-  CC_CHECK(hipMalloc(&dC, elems * sizeof(float)));
-  CC_CHECK(hipMalloc(&dRef, elems * sizeof(float)));
-  CC_CHECK(hipMalloc(&dErr, sizeof(float)));
-  CC_CHECK(hipMalloc(&dSum, sizeof(unsigned long long)));
+  std::lock_guard<std::mutex> lk(g_ctx_mu);
+  if (device < 0 || device >= kMaxDevices) return -5;
+  ProbeCtx* ctx = nullptr;
+  CC_CHECK(ctx_acquire(device, D, &ctx));
+  bf16 *dA = ctx->dA, *dB = ctx->dB;
+  float *dC = ctx->dC, *dRef = ctx->dRef, *dErr = ctx->dErr;
+  unsigned long long* dSum = ctx->dSum;
+  hipEvent_t ev0 = ctx->ev0, ev1 = ctx->ev1;
   CC_CHECK(hipMemset(dErr, 0, sizeof(float)));
   CC_CHECK(hipMemset(dSum, 0, sizeof(unsigned long long)));
-
-  hipEvent_t ev0, ev1;
-  CC_CHECK(hipEventCreate(&ev0));
-  CC_CHECK(hipEventCreate(&ev1));
 
   // -- fill (asymmetric seeds: catches row/col-swapped layouts) --------
   hipLaunchKernelGGL(fill_bf16_lcg, dim3(2048), dim3(256), 0, 0, dA, elems, 1u);
@@ -434,8 +526,7 @@ int cc_attest_device(int device, int gemm_dim, struct CcAttestReport* rep) {
                      hipMemcpyDeviceToHost));
 
   // -- LDS probe -------------------------------------------------------
-  uint32_t* dFail = nullptr;
-  CC_CHECK(hipMalloc(&dFail, sizeof(uint32_t)));
+  uint32_t* dFail = ctx->dFail;
   CC_CHECK(hipMemset(dFail, 0, sizeof(uint32_t)));
   CC_CHECK(hipEventRecord(ev0, 0));
   hipLaunchKernelGGL(lds_probe, dim3(512), dim3(256), 0, 0, dFail, 8);
@@ -444,7 +535,6 @@ int cc_attest_device(int device, int gemm_dim, struct CcAttestReport* rep) {
   rep->lds_ms = event_ms(ev0, ev1);
   CC_CHECK(hipMemcpy(&rep->lds_failures, dFail, sizeof(uint32_t),
                      hipMemcpyDeviceToHost));
-  hipFree(dFail);
 
   // -- HBM probe: reuse C/Ref buffers as src/dst -----------------------
   long n4 = elems / 4;
@@ -467,14 +557,20 @@ int cc_attest_device(int device, int gemm_dim, struct CcAttestReport* rep) {
       ++rep->peers_accessible;
   }
 
-  hipEventDestroy(ev0);
-  hipEventDestroy(ev1);
-  hipFree(dA); hipFree(dB); hipFree(dC); hipFree(dRef);
-  hipFree(dErr); hipFree(dSum);
-
   rep->ok = (rep->max_abs_err == 0.0f) && (rep->lds_failures == 0) &&
             (rep->gemm_tflops > 0.0) ? 1 : 0;
   return 0;
+}
+
+// Free every cached probe context (daemon shutdown / tests).
+void cc_attest_shutdown(void) {
+  std::lock_guard<std::mutex> lk(g_ctx_mu);
+  for (int i = 0; i < kMaxDevices; ++i) {
+    if (g_ctx[i].dim || g_ctx[i].dLive) {
+      (void)hipSetDevice(i);
+      ctx_release(g_ctx[i]);
+    }
+  }
 }
 
 }  // extern "C"

@@ -1,0 +1,49 @@
+#!/usr/bin/env python3
+"""MFMA GEMM perf sweep on the attestation kernel (run on a GPU box).
+
+Reports TF/s at several sizes using torch-allocated buffers, validates
+against torch fp32 at the smallest size, and prints one JSON line.
+"""
+
+import json
+import sys
+import time
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+
+import torch
+
+from k8s_cc_manager_amd.ops import attest
+
+
+def bench_size(n: int, iters: int = 20) -> dict:
+    a = torch.randn(n, n, device="cuda", dtype=torch.float32).bfloat16()
+    bt = torch.randn(n, n, device="cuda", dtype=torch.float32).bfloat16()
+    c = torch.empty(n, n, device="cuda", dtype=torch.float32)
+    # warm
+    attest.mfma_gemm_bf16(0, a.data_ptr(), bt.data_ptr(), c.data_ptr(), n, n, n)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        attest.mfma_gemm_bf16(0, a.data_ptr(), bt.data_ptr(), c.data_ptr(), n, n, n)
+    torch.cuda.synchronize()
+    dt = (time.perf_counter() - t0) / iters
+    tf = 2.0 * n * n * n / dt / 1e12
+    out = {"n": n, "ms": round(dt * 1e3, 3), "tflops": round(tf, 1)}
+    if n <= 1024:
+        ref = a.float() @ bt.float().t()
+        out["max_abs_err"] = round((c - ref).abs().max().item(), 6)
+    del a, bt, c
+    torch.cuda.empty_cache()
+    return out
+
+
+def main():
+    sizes = [int(s) for s in sys.argv[1:]] or [1024, 2048, 4096, 8192]
+    results = [bench_size(n) for n in sizes]
+    print(json.dumps({"kernel": "mfma_gemm_bf16", "results": results}))
+
+
+if __name__ == "__main__":
+    main()
