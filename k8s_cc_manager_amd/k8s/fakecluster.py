@@ -214,6 +214,8 @@ class FakeCluster:
 
         class Handler(BaseHTTPRequestHandler):
             protocol_version = "HTTP/1.1"
+            # Nagle + delayed-ACK costs ~40 ms per keep-alive round trip
+            disable_nagle_algorithm = True
 
             def log_message(self, fmt: str, *args: Any) -> None:  # quiet
                 logger.debug("fakeapi: " + fmt, *args)
