@@ -30,6 +30,7 @@ from ..labels import (
     STATE_FAILED,
     VALID_MODES,
 )
+from ..utils import eventlog
 from ..utils.metrics import METRICS
 from ..utils.readiness import create_readiness_file
 from .transition import TransitionEngine, TransitionReport
@@ -200,6 +201,15 @@ class CCManager:
         report: TransitionReport = runner()
         self.last_report = report
         METRICS.observe_transition(mode, report.ok, report.seconds, report.phases)
+        eventlog.record_transition(
+            self.node_name,
+            mode,
+            report.ok,
+            report.seconds,
+            phases=report.phases,
+            devices_changed=report.devices_changed,
+            error=report.error,
+        )
         self._set_state(mode if report.ok else STATE_FAILED)
         return report.ok
 

@@ -130,6 +130,10 @@ def _load() -> ctypes.CDLL:
     ]
     lib.cc_ref_gemm_f32.restype = ctypes.c_int
     lib.cc_ref_gemm_f32.argtypes = lib.cc_mfma_gemm_bf16.argtypes
+    lib.cc_mfma_gemm_bf16_variant.restype = ctypes.c_int
+    lib.cc_mfma_gemm_bf16_variant.argtypes = lib.cc_mfma_gemm_bf16.argtypes + [
+        ctypes.c_int
+    ]
     _lib = lib
     return lib
 
@@ -226,3 +230,14 @@ def ref_gemm_f32(device_index: int, a_ptr: int, bt_ptr: int, c_ptr: int,
     rc = _load().cc_ref_gemm_f32(device_index, a_ptr, bt_ptr, c_ptr, m, n, k)
     if rc != 0:
         raise AttestationError(f"ref_gemm_f32 rc={rc}")
+
+
+def mfma_gemm_bf16_variant(device_index: int, a_ptr: int, bt_ptr: int,
+                           c_ptr: int, m: int, n: int, k: int,
+                           which: int) -> None:
+    """Force a GEMM variant: 0 = 128x128 step-3, 1 = 256x256 8-phase."""
+    rc = _load().cc_mfma_gemm_bf16_variant(
+        device_index, a_ptr, bt_ptr, c_ptr, m, n, k, which
+    )
+    if rc != 0:
+        raise AttestationError(f"mfma_gemm_bf16_variant({which}) rc={rc}")

@@ -197,6 +197,178 @@ __global__ __launch_bounds__(256, 2) void mfma_gemm_bf16(
 }
 
 // ---------------------------------------------------------------------------
+// 256x256 8-phase deep-pipelined MFMA GEMM (the CDNA4 guide's 256²
+// template class). 512 threads = 8 waves (2M x 4N), BK=64, 128 KiB LDS:
+// per double-buffer slot one K-tile of A and B, each in two 16 KiB
+// halves (A rows 0-127 / 128-255; Bt rows likewise). Each phase
+// computes one 128x128 block-quadrant (mh,nh) x K=64 — per wave 16
+// MFMA — touching ONLY A-half mh and B-half nh, so half-tile slots free
+// progressively and one half-tile is prefetched per phase with raw
+// s_barrier + counted s_waitcnt vmcnt (never vmcnt(0) mid-loop except
+// the final two iterations, where prefetch guards would otherwise leave
+// needed DMAs undrained).
+//
+// Phase schedule per iteration J (t = 2J; quadrant order A0B0, A0B1,
+// A1B1, A1B0 for tile t in buf0 then tile t+1 in buf1):
+//   p1 computes t:A0B0, prefetches A1(t+1); p2 t:A0B1, B0(t+1);
+//   p3 t:A1B1, A0(t+2); p4 t:A1B0, B1(t+2) + vmcnt(4) + barrier;
+//   p5 t+1:A0B0, A1(t+2); p6 t+1:A0B1, B0(t+2);
+//   p7 t+1:A1B1, A0(t+3); p8 t+1:A1B0, B1(t+3) + vmcnt(4) + barrier.
+// With 2 glds per wave per phase, vmcnt(4)+s_barrier guarantees every
+// wave's DMAs older than two phases are visible chip-wide before any
+// consumer ds_read — each needed half-tile is staged >= 3 phases before
+// first read (verified bitwise by the integer attest data, and against
+// torch fp32 in tests/test_gpu_attest.py).
+// ---------------------------------------------------------------------------
+constexpr int BM2 = 256, BN2 = 256, BK2 = 64;
+constexpr int HALF_B = 16384;  // one half-tile (128 rows x 128 B)
+
+__device__ __forceinline__ char* slot_ptr(char* lds, int op, int buf, int half) {
+  return lds + ((op * 2 + buf) * 2 + half) * HALF_B;
+}
+
+// Stage one 16 KiB half-tile with 8 waves x 2 glds (1 KiB per wave per
+// instruction); st_16x32 swizzle rides the global source address.
+__device__ __forceinline__ void stage_half_glds(
+    const char* gbase, long row_stride_b, long k0_b, int row0,
+    char* lds_half, int wave, int lane) {
+#pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    int base = (p * 8 + wave) * 1024;
+    int logical = swz(base + lane * 16);
+    int row = logical >> 7;
+    int colb = logical & 127;
+    const char* g = gbase + (long)(row0 + row) * row_stride_b + k0_b + colb;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)g,
+        (__attribute__((address_space(3))) void*)(lds_half + base), 16, 0, 0);
+  }
+}
+
+struct Gemm256Args {
+  const char* gA;
+  const char* gB;
+  long row_b;
+  int nk;
+};
+
+// one phase: ds-read quadrant fragments, issue one half-tile prefetch,
+// raw barrier, lgkmcnt(0), 16 MFMA at prio 1, raw barrier.
+#define PHASE(buf, mh, nh, ACC, PREFETCH_STMT, DRAIN)                          \
+  do {                                                                         \
+    char* Ah = slot_ptr(lds, 0, (buf), (mh));                                  \
+    char* Bh = slot_ptr(lds, 1, (buf), (nh));                                  \
+    bf16x8 af[4][2], bf[2][2];                                                 \
+    _Pragma("unroll") for (int ks = 0; ks < 2; ++ks) {                         \
+      _Pragma("unroll") for (int i = 0; i < 4; ++i) {                          \
+        int lg = (wave_mq + i * 16 + lane15) * 128 + ks * 64 + khalf_b;        \
+        af[i][ks] = *(const bf16x8*)(Ah + swz(lg));                            \
+      }                                                                        \
+      _Pragma("unroll") for (int j = 0; j < 2; ++j) {                          \
+        int lg = (wave_nq + j * 16 + lane15) * 128 + ks * 64 + khalf_b;        \
+        bf[j][ks] = *(const bf16x8*)(Bh + swz(lg));                            \
+      }                                                                        \
+    }                                                                          \
+    PREFETCH_STMT;                                                             \
+    DRAIN;                                                                     \
+    __builtin_amdgcn_s_barrier();                                              \
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                         \
+    __builtin_amdgcn_s_setprio(1);                                             \
+    _Pragma("unroll") for (int ks = 0; ks < 2; ++ks)                           \
+        _Pragma("unroll") for (int i = 0; i < 4; ++i)                          \
+        _Pragma("unroll") for (int j = 0; j < 2; ++j)                          \
+            ACC[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(               \
+                af[i][ks], bf[j][ks], ACC[i][j], 0, 0, 0);                     \
+    __builtin_amdgcn_s_setprio(0);                                             \
+    __builtin_amdgcn_s_barrier();                                              \
+  } while (0)
+
+__global__ __launch_bounds__(512, 1) void mfma_gemm_bf16_256(
+    const bf16* __restrict__ A, const bf16* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K) {
+  __shared__ char lds[8 * HALF_B];  // 128 KiB, ONE shared object
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;               // 0..7
+  const int wave_mq = (wave >> 2) * 64;    // row offset inside a quadrant
+  const int wave_nq = (wave & 3) * 32;     // col offset inside a quadrant
+  const int lane15 = lane & 15;
+  const int khalf_b = (lane >> 4) * 16;
+  const int block_m = blockIdx.y * BM2;
+  const int block_n = blockIdx.x * BN2;
+
+  const char* gA = (const char*)(A + (long)block_m * K);
+  const char* gB = (const char*)(Bt + (long)block_n * K);
+  const long row_b = (long)K * 2;
+  const int nk = K / BK2;
+
+  // acc[mh][nh][i][j]: 2x2 quadrants x (4 m-frag x 2 n-frag) = 128 VGPR
+  f32x4 acc00[4][2] = {}, acc01[4][2] = {}, acc10[4][2] = {}, acc11[4][2] = {};
+
+#define STAGE(op, buf, half, tile)                                             \
+  stage_half_glds((op) == 0 ? gA : gB, row_b, (long)(tile) * BK2 * 2,          \
+                  (half) * 128, slot_ptr(lds, (op), (buf), (half)), wave, lane)
+
+  // prologue: tile 0 complete + A0,B1 of tile 1 (the halves iteration 0
+  // does not stage itself); full drain once.
+  STAGE(0, 0, 0, 0);
+  STAGE(1, 0, 0, 0);
+  STAGE(0, 0, 1, 0);
+  STAGE(1, 0, 1, 0);
+  STAGE(0, 1, 0, 1);
+  STAGE(1, 1, 1, 1);
+  __syncthreads();
+
+#define VM_DRAIN                                                               \
+  do {                                                                         \
+    if (tp + 4 >= nk)                                                          \
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");                         \
+    else                                                                       \
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");                         \
+  } while (0)
+
+  for (int tp = 0; tp < nk; tp += 2) {
+    // tile tp from buf0 ------------------------------------------------
+    PHASE(0, 0, 0, acc00, if (tp + 1 < nk) STAGE(0, 1, 1, tp + 1), );
+    PHASE(0, 0, 1, acc01, if (tp + 1 < nk) STAGE(1, 1, 0, tp + 1), );
+    PHASE(0, 1, 1, acc11, if (tp + 2 < nk) STAGE(0, 0, 0, tp + 2), );
+    PHASE(0, 1, 0, acc10, if (tp + 2 < nk) STAGE(1, 0, 1, tp + 2), VM_DRAIN);
+    // tile tp+1 from buf1 ----------------------------------------------
+    PHASE(1, 0, 0, acc00, if (tp + 2 < nk) STAGE(0, 0, 1, tp + 2), );
+    PHASE(1, 0, 1, acc01, if (tp + 2 < nk) STAGE(1, 0, 0, tp + 2), );
+    PHASE(1, 1, 1, acc11, if (tp + 3 < nk) STAGE(0, 1, 0, tp + 3), );
+    PHASE(1, 1, 0, acc10, if (tp + 3 < nk) STAGE(1, 1, 1, tp + 3), VM_DRAIN);
+  }
+#undef VM_DRAIN
+#undef STAGE
+
+  // epilogue: C/D mapping col=lane&15, row=(lane>>4)*4+r
+  const int c_col = lane & 15;
+  const int c_row0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int mh = 0; mh < 2; ++mh) {
+#pragma unroll
+    for (int nh = 0; nh < 2; ++nh) {
+      f32x4(*acc)[2] = mh == 0 ? (nh == 0 ? acc00 : acc01)
+                               : (nh == 0 ? acc10 : acc11);
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+#pragma unroll
+        for (int j = 0; j < 2; ++j) {
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            int row = block_m + mh * 128 + wave_mq + i * 16 + c_row0 + r;
+            int col = block_n + nh * 128 + wave_nq + j * 16 + c_col;
+            C[(long)row * N + col] = acc[i][j][r];
+          }
+        }
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // VALU fp32 reference GEMM (independent ground truth; deliberately does
 // NOT share tiling or fragment code with the MFMA path).
 // One thread per C element, fp32 FMA chain over K.
@@ -432,15 +604,50 @@ int cc_device_index_for_bdf(int domain, int bus, int dev) {
   return -1;
 }
 
-// Run the bf16 MFMA GEMM on caller-provided device buffers (e.g. torch
-// tensors): C[M,N] = A[M,K] @ Bt[N,K]^T. M,N multiples of 128, K of 32.
-int cc_mfma_gemm_bf16(int device, const void* A, const void* Bt, void* C,
-                      int M, int N, int K) {
+// Launch the best-fitting MFMA GEMM variant (no sync).
+static int launch_mfma_gemm(const void* A, const void* Bt, void* C, int M,
+                            int N, int K) {
+  if (M % BM2 == 0 && N % BN2 == 0 && K % (2 * BK2) == 0) {
+    dim3 grid(N / BN2, M / BM2);
+    hipLaunchKernelGGL(mfma_gemm_bf16_256, grid, dim3(512), 0, 0,
+                       (const bf16*)A, (const bf16*)Bt, (float*)C, M, N, K);
+    return 0;
+  }
   if (M % BM || N % BN || K % BK) return -2;
-  if (hipSetDevice(device) != hipSuccess) return -3;
   dim3 grid(N / BN, M / BM);
   hipLaunchKernelGGL(mfma_gemm_bf16, grid, dim3(256), 0, 0, (const bf16*)A,
                      (const bf16*)Bt, (float*)C, M, N, K);
+  return 0;
+}
+
+// Run the bf16 MFMA GEMM on caller-provided device buffers (e.g. torch
+// tensors): C[M,N] = A[M,K] @ Bt[N,K]^T. M,N multiples of 128, K of 32
+// (the deep-pipelined 256x256 variant is selected automatically when
+// M,N are multiples of 256 and K of 128).
+int cc_mfma_gemm_bf16(int device, const void* A, const void* Bt, void* C,
+                      int M, int N, int K) {
+  if (hipSetDevice(device) != hipSuccess) return -3;
+  int rc = launch_mfma_gemm(A, Bt, C, M, N, K);
+  if (rc != 0) return rc;
+  return (int)hipDeviceSynchronize();
+}
+
+// Force a specific variant (perf characterization / A-B): which=0 the
+// 128x128 step-3 kernel, which=1 the 256x256 8-phase kernel.
+int cc_mfma_gemm_bf16_variant(int device, const void* A, const void* Bt,
+                              void* C, int M, int N, int K, int which) {
+  if (hipSetDevice(device) != hipSuccess) return -3;
+  if (which == 1) {
+    if (M % BM2 || N % BN2 || K % (2 * BK2)) return -2;
+    dim3 grid(N / BN2, M / BM2);
+    hipLaunchKernelGGL(mfma_gemm_bf16_256, grid, dim3(512), 0, 0,
+                       (const bf16*)A, (const bf16*)Bt, (float*)C, M, N, K);
+  } else {
+    if (M % BM || N % BN || K % BK) return -2;
+    dim3 grid(N / BN, M / BM);
+    hipLaunchKernelGGL(mfma_gemm_bf16, grid, dim3(256), 0, 0, (const bf16*)A,
+                       (const bf16*)Bt, (float*)C, M, N, K);
+  }
   return (int)hipDeviceSynchronize();
 }
 
@@ -492,12 +699,11 @@ int cc_attest_device(int device, int gemm_dim, struct CcAttestReport* rep) {
   hipLaunchKernelGGL(fill_bf16_lcg, dim3(2048), dim3(256), 0, 0, dA, elems, 1u);
   hipLaunchKernelGGL(fill_bf16_lcg, dim3(2048), dim3(256), 0, 0, dB, elems, 7u);
 
-  // -- MFMA GEMM (timed; warm once) ------------------------------------
-  dim3 grid(D / BN, D / BM);
-  hipLaunchKernelGGL(mfma_gemm_bf16, grid, dim3(256), 0, 0, dA, dB, dC, D, D, D);
+  // -- MFMA GEMM (timed; warm once; best-fitting variant) --------------
+  launch_mfma_gemm(dA, dB, dC, D, D, D);
   CC_CHECK(hipDeviceSynchronize());
   CC_CHECK(hipEventRecord(ev0, 0));
-  hipLaunchKernelGGL(mfma_gemm_bf16, grid, dim3(256), 0, 0, dA, dB, dC, D, D, D);
+  launch_mfma_gemm(dA, dB, dC, D, D, D);
   CC_CHECK(hipEventRecord(ev1, 0));
   CC_CHECK(hipEventSynchronize(ev1));
   rep->gemm_ms = event_ms(ev0, ev1);

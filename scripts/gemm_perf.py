@@ -21,19 +21,26 @@ def bench_size(n: int, iters: int = 20) -> dict:
     a = torch.randn(n, n, device="cuda", dtype=torch.float32).bfloat16()
     bt = torch.randn(n, n, device="cuda", dtype=torch.float32).bfloat16()
     c = torch.empty(n, n, device="cuda", dtype=torch.float32)
-    # warm
-    attest.mfma_gemm_bf16(0, a.data_ptr(), bt.data_ptr(), c.data_ptr(), n, n, n)
-    torch.cuda.synchronize()
-    t0 = time.perf_counter()
-    for _ in range(iters):
-        attest.mfma_gemm_bf16(0, a.data_ptr(), bt.data_ptr(), c.data_ptr(), n, n, n)
-    torch.cuda.synchronize()
-    dt = (time.perf_counter() - t0) / iters
-    tf = 2.0 * n * n * n / dt / 1e12
-    out = {"n": n, "ms": round(dt * 1e3, 3), "tflops": round(tf, 1)}
-    if n <= 1024:
-        ref = a.float() @ bt.float().t()
-        out["max_abs_err"] = round((c - ref).abs().max().item(), 6)
+    out = {"n": n}
+    for which, name in ((0, "v128"), (1, "v256")):
+        if which == 1 and (n % 256 or n % 128):
+            continue
+        run = lambda: attest.mfma_gemm_bf16_variant(  # noqa: E731
+            0, a.data_ptr(), bt.data_ptr(), c.data_ptr(), n, n, n, which
+        )
+        run()  # warm + correctness
+        torch.cuda.synchronize()
+        if n <= 2048:
+            ref = a.float() @ bt.float().t()
+            err = (c - ref).abs().max().item()
+            out[f"{name}_max_abs_err"] = round(err, 6)
+        t0 = time.perf_counter()
+        for _ in range(iters):
+            run()
+        torch.cuda.synchronize()
+        dt = (time.perf_counter() - t0) / iters
+        out[f"{name}_ms"] = round(dt * 1e3, 3)
+        out[f"{name}_tflops"] = round(2.0 * n * n * n / dt / 1e12, 1)
     del a, bt, c
     torch.cuda.empty_cache()
     return out
@@ -42,7 +49,7 @@ def bench_size(n: int, iters: int = 20) -> dict:
 def main():
     sizes = [int(s) for s in sys.argv[1:]] or [1024, 2048, 4096, 8192]
     results = [bench_size(n) for n in sizes]
-    print(json.dumps({"kernel": "mfma_gemm_bf16", "results": results}))
+    print(json.dumps({"kernel": "mfma_gemm_bf16 A/B", "results": results}))
 
 
 if __name__ == "__main__":

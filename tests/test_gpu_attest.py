@@ -60,6 +60,41 @@ def test_mfma_gemm_integer_exact(attest):
     assert torch.equal(c, ref)
 
 
+@pytest.mark.parametrize("m,n,k", [(256, 256, 128), (512, 256, 384), (256, 512, 1024)])
+def test_mfma_gemm_256_template_matches_torch(attest, m, n, k):
+    """The deep-pipelined 256x256 8-phase variant — race-sensitive, so
+    run it several times per shape (sync-structure screening)."""
+    torch.manual_seed(m * 7 + n * 3 + k)
+    a = torch.randn(m, k, device="cuda", dtype=torch.float32).bfloat16()
+    bt = torch.randn(n, k, device="cuda", dtype=torch.float32).bfloat16()
+    ref = a.float() @ bt.float().t()
+    scale = ref.abs().max().item()
+    for trial in range(5):
+        c = torch.full((m, n), float("nan"), device="cuda", dtype=torch.float32)
+        attest.mfma_gemm_bf16_variant(
+            0, a.data_ptr(), bt.data_ptr(), c.data_ptr(), m, n, k, 1
+        )
+        torch.cuda.synchronize()
+        err = (c - ref).abs().max().item()
+        assert err <= 2e-3 * max(scale, 1.0), f"trial {trial}: err={err}"
+
+
+def test_mfma_gemm_256_integer_exact_race_screen(attest):
+    """Integer data: any stale-LDS race shows as a bitwise mismatch."""
+    m = n = 512
+    k = 768
+    a = (torch.randint(-2, 2, (m, k), device="cuda")).bfloat16()
+    bt = (torch.randint(-2, 2, (n, k), device="cuda")).bfloat16()
+    ref = a.float() @ bt.float().t()
+    for trial in range(10):
+        c = torch.empty(m, n, device="cuda", dtype=torch.float32)
+        attest.mfma_gemm_bf16_variant(
+            0, a.data_ptr(), bt.data_ptr(), c.data_ptr(), m, n, k, 1
+        )
+        torch.cuda.synchronize()
+        assert torch.equal(c, ref), f"trial {trial} mismatch"
+
+
 def test_ref_gemm_matches_torch(attest):
     m, n, k = 128, 128, 256
     a = torch.randn(m, k, device="cuda").bfloat16()
