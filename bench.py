@@ -45,6 +45,13 @@ def parse_args(argv=None):
     p.add_argument("--attest-dim", type=int, default=1024, help="attestation GEMM size")
     p.add_argument("--mock", action="store_true", help="force the mock device tier")
     p.add_argument("--no-evict", action="store_true", help="skip the eviction leg")
+    p.add_argument(
+        "--workload",
+        action="store_true",
+        help="run a synthetic HIP workload (continuous MFMA GEMMs) during "
+        "the rolling toggle — BASELINE config-5 analogue (evict + readmit "
+        "under load)",
+    )
     p.add_argument("--json-out", default="", help="also write the JSON line here")
     return p.parse_args(argv)
 
@@ -124,6 +131,29 @@ def main(argv=None) -> int:
         ),
     )
 
+    # ---- optional synthetic workload (config 5: toggle under load) ----
+    workload_stop = None
+    workload_thread = None
+    if args.workload and use_gpu:
+        import threading
+
+        from k8s_cc_manager_amd.ops import attest as _att
+
+        workload_stop = threading.Event()
+        wa = torch.randn(512, 512, device="cuda").bfloat16()
+        wb = torch.randn(512, 512, device="cuda").bfloat16()
+        wc = torch.empty(512, 512, device="cuda", dtype=torch.float32)
+
+        def _workload():
+            while not workload_stop.is_set():
+                _att.mfma_gemm_bf16(
+                    indices[0], wa.data_ptr(), wb.data_ptr(), wc.data_ptr(),
+                    512, 512, 512,
+                )
+
+        workload_thread = threading.Thread(target=_workload, daemon=True)
+        workload_thread.start()
+
     def reconcile_step(i: int) -> None:
         mode = "on" if i % 2 == 0 else "off"
         cluster.set_node_label(node_name, CC_MODE_LABEL, mode)
@@ -186,6 +216,7 @@ def main(argv=None) -> int:
                 "eviction": not args.no_evict,
                 "components": len(COMPONENT_LABELS),
                 "attest_gemm_dim": args.attest_dim if use_gpu else 0,
+                "workload": bool(args.workload and use_gpu),
                 "sec_per_gpu_transition": round(elapsed / args.steps, 4),
             },
         }
@@ -195,6 +226,9 @@ def main(argv=None) -> int:
             with open(args.json_out, "w") as f:
                 f.write(line + "\n")
 
+    if workload_stop is not None:
+        workload_stop.set()
+        workload_thread.join(timeout=10)
     cluster.stop()
     if world > 1:
         dist.destroy_process_group()

@@ -252,19 +252,25 @@ struct Gemm256Args {
   int nk;
 };
 
-// one phase: ds-read quadrant fragments, issue one half-tile prefetch,
+// one phase: ds-read only the fragments this quadrant does NOT already
+// hold (adjacent phases share an A-half or a B-half of the same K-tile,
+// so those fragments persist in VGPRs across the phase barrier: phases
+// read 12/4/8/4 b128 instead of 12 each), issue one half-tile prefetch,
 // raw barrier, lgkmcnt(0), 16 MFMA at prio 1, raw barrier.
-#define PHASE(buf, mh, nh, ACC, PREFETCH_STMT, DRAIN)                          \
+#define PHASE(buf, mh, nh, LOAD_A, LOAD_B, ACC, PREFETCH_STMT, DRAIN)          \
   do {                                                                         \
-    char* Ah = slot_ptr(lds, 0, (buf), (mh));                                  \
-    char* Bh = slot_ptr(lds, 1, (buf), (nh));                                  \
-    bf16x8 af[4][2], bf[2][2];                                                 \
-    _Pragma("unroll") for (int ks = 0; ks < 2; ++ks) {                         \
-      _Pragma("unroll") for (int i = 0; i < 4; ++i) {                          \
+    if (LOAD_A) {                                                              \
+      char* Ah = slot_ptr(lds, 0, (buf), (mh));                                \
+      _Pragma("unroll") for (int ks = 0; ks < 2; ++ks)                         \
+          _Pragma("unroll") for (int i = 0; i < 4; ++i) {                      \
         int lg = (wave_mq + i * 16 + lane15) * 128 + ks * 64 + khalf_b;        \
         af[i][ks] = *(const bf16x8*)(Ah + swz(lg));                            \
       }                                                                        \
-      _Pragma("unroll") for (int j = 0; j < 2; ++j) {                          \
+    }                                                                          \
+    if (LOAD_B) {                                                              \
+      char* Bh = slot_ptr(lds, 1, (buf), (nh));                                \
+      _Pragma("unroll") for (int ks = 0; ks < 2; ++ks)                         \
+          _Pragma("unroll") for (int j = 0; j < 2; ++j) {                      \
         int lg = (wave_nq + j * 16 + lane15) * 128 + ks * 64 + khalf_b;        \
         bf[j][ks] = *(const bf16x8*)(Bh + swz(lg));                            \
       }                                                                        \
@@ -305,6 +311,9 @@ __global__ __launch_bounds__(512, 1) void mfma_gemm_bf16_256(
 
   // acc[mh][nh][i][j]: 2x2 quadrants x (4 m-frag x 2 n-frag) = 128 VGPR
   f32x4 acc00[4][2] = {}, acc01[4][2] = {}, acc10[4][2] = {}, acc11[4][2] = {};
+  // fragment registers persist across phases (A shared by quadrant
+  // pairs (mh,0)/(mh,1); B1 shared by phases 2-3, same K-tile)
+  bf16x8 af[4][2], bf[2][2];
 
 #define STAGE(op, buf, half, tile)                                             \
   stage_half_glds((op) == 0 ? gA : gB, row_b, (long)(tile) * BK2 * 2,          \
@@ -329,16 +338,16 @@ __global__ __launch_bounds__(512, 1) void mfma_gemm_bf16_256(
   } while (0)
 
   for (int tp = 0; tp < nk; tp += 2) {
-    // tile tp from buf0 ------------------------------------------------
-    PHASE(0, 0, 0, acc00, if (tp + 1 < nk) STAGE(0, 1, 1, tp + 1), );
-    PHASE(0, 0, 1, acc01, if (tp + 1 < nk) STAGE(1, 1, 0, tp + 1), );
-    PHASE(0, 1, 1, acc11, if (tp + 2 < nk) STAGE(0, 0, 0, tp + 2), );
-    PHASE(0, 1, 0, acc10, if (tp + 2 < nk) STAGE(1, 0, 1, tp + 2), VM_DRAIN);
+    // tile tp from buf0 (reads: 12 / 4 / 8 / 4 ds_read_b128) -----------
+    PHASE(0, 0, 0, 1, 1, acc00, if (tp + 1 < nk) STAGE(0, 1, 1, tp + 1), );
+    PHASE(0, 0, 1, 0, 1, acc01, if (tp + 1 < nk) STAGE(1, 1, 0, tp + 1), );
+    PHASE(0, 1, 1, 1, 0, acc11, if (tp + 2 < nk) STAGE(0, 0, 0, tp + 2), );
+    PHASE(0, 1, 0, 0, 1, acc10, if (tp + 2 < nk) STAGE(1, 0, 1, tp + 2), VM_DRAIN);
     // tile tp+1 from buf1 ----------------------------------------------
-    PHASE(1, 0, 0, acc00, if (tp + 2 < nk) STAGE(0, 0, 1, tp + 2), );
-    PHASE(1, 0, 1, acc01, if (tp + 2 < nk) STAGE(1, 0, 0, tp + 2), );
-    PHASE(1, 1, 1, acc11, if (tp + 3 < nk) STAGE(0, 1, 0, tp + 3), );
-    PHASE(1, 1, 0, acc10, if (tp + 3 < nk) STAGE(1, 1, 1, tp + 3), VM_DRAIN);
+    PHASE(1, 0, 0, 1, 1, acc00, if (tp + 2 < nk) STAGE(0, 0, 1, tp + 2), );
+    PHASE(1, 0, 1, 0, 1, acc01, if (tp + 2 < nk) STAGE(1, 0, 0, tp + 2), );
+    PHASE(1, 1, 1, 1, 0, acc11, if (tp + 3 < nk) STAGE(0, 1, 0, tp + 3), );
+    PHASE(1, 1, 0, 0, 1, acc10, if (tp + 3 < nk) STAGE(1, 1, 1, tp + 3), VM_DRAIN);
   }
 #undef VM_DRAIN
 #undef STAGE

@@ -13,6 +13,12 @@ def pytest_configure(config):
     )
 
 
+@pytest.fixture(autouse=True)
+def _isolated_event_log(tmp_path, monkeypatch):
+    """Keep transition event logs out of /var/lib during tests."""
+    monkeypatch.setenv("CC_EVENT_LOG", str(tmp_path / "transitions.jsonl"))
+
+
 @pytest.fixture
 def fake_cluster():
     from k8s_cc_manager_amd.k8s.fakecluster import FakeCluster
