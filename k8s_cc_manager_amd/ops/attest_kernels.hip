@@ -209,16 +209,20 @@ __global__ __launch_bounds__(256, 2) void mfma_gemm_bf16(
 // needed DMAs undrained).
 //
 // Phase schedule per iteration J (t = 2J; quadrant order A0B0, A0B1,
-// A1B1, A1B0 for tile t in buf0 then tile t+1 in buf1):
-//   p1 computes t:A0B0, prefetches A1(t+1); p2 t:A0B1, B0(t+1);
-//   p3 t:A1B1, A0(t+2); p4 t:A1B0, B1(t+2) + vmcnt(4) + barrier;
-//   p5 t+1:A0B0, A1(t+2); p6 t+1:A0B1, B0(t+2);
-//   p7 t+1:A1B1, A0(t+3); p8 t+1:A1B0, B1(t+3) + vmcnt(4) + barrier.
-// With 2 glds per wave per phase, vmcnt(4)+s_barrier guarantees every
-// wave's DMAs older than two phases are visible chip-wide before any
-// consumer ds_read — each needed half-tile is staged >= 3 phases before
-// first read (verified bitwise by the integer attest data, and against
-// torch fp32 in tests/test_gpu_attest.py).
+// A1B1, A1B0 for tile t in buf0 then tile t+1 in buf1; A fragments are
+// register-reused by quadrant pairs and B1 by phases 2-3, so the LDS
+// read phases per slot are: A0@p1, B0@p1+p4, B1@p2, A1@p3 — slots free
+// progressively and each phase prefetches exactly the slot freed the
+// phase before):
+//   p1 t:A0B0, prefetch B0(t+1);  p2 t:A0B1, A0(t+2);
+//   p3 t:A1B1, B1(t+2);           p4 t:A1B0, A1(t+2) + vmcnt(6);
+//   p5 t+1:A0B0, B0(t+2);         p6 t+1:A0B1, A0(t+3);
+//   p7 t+1:A1B1, B1(t+3);         p8 t+1:A1B0, A1(t+3) + vmcnt(6).
+// At each vmcnt(6)+s_barrier the three newest half-tiles may still fly;
+// everything a following phase reads is >= 3 phases old and therefore
+// drained for EVERY wave (the barrier makes the per-wave vmcnt
+// chip-wide). Race-screened bitwise on integer data and against torch
+// fp32 (tests/test_gpu_attest.py).
 // ---------------------------------------------------------------------------
 constexpr int BM2 = 256, BN2 = 256, BK2 = 64;
 constexpr int HALF_B = 16384;  // one half-tile (128 rows x 128 B)
@@ -291,7 +295,7 @@ struct Gemm256Args {
 
 __global__ __launch_bounds__(512, 1) void mfma_gemm_bf16_256(
     const bf16* __restrict__ A, const bf16* __restrict__ Bt,
-    float* __restrict__ C, int M, int N, int K) {
+    float* __restrict__ C, int M, int N, int K, int xcd_swizzle) {
   __shared__ char lds[8 * HALF_B];  // 128 KiB, ONE shared object
 
   const int tid = threadIdx.x;
@@ -301,8 +305,18 @@ __global__ __launch_bounds__(512, 1) void mfma_gemm_bf16_256(
   const int wave_nq = (wave & 3) * 32;     // col offset inside a quadrant
   const int lane15 = lane & 15;
   const int khalf_b = (lane >> 4) * 16;
-  const int block_m = blockIdx.y * BM2;
-  const int block_n = blockIdx.x * BN2;
+  // XCD-aware bijective remap (dispatcher places block b on XCD b%8):
+  // consecutive remapped ids share an XCD's L2 -> tile locality when
+  // HBM-bound; host enables it only when the working set exceeds L3.
+  int wg = blockIdx.y * gridDim.x + blockIdx.x;
+  if (xcd_swizzle) {
+    int nwg = gridDim.x * gridDim.y;
+    int q = nwg >> 3, r = nwg & 7;
+    int xcd = wg & 7, o = wg >> 3;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + o;
+  }
+  const int block_m = (wg / gridDim.x) * BM2;
+  const int block_n = (wg % gridDim.x) * BN2;
 
   const char* gA = (const char*)(A + (long)block_m * K);
   const char* gB = (const char*)(Bt + (long)block_n * K);
@@ -319,14 +333,15 @@ __global__ __launch_bounds__(512, 1) void mfma_gemm_bf16_256(
   stage_half_glds((op) == 0 ? gA : gB, row_b, (long)(tile) * BK2 * 2,          \
                   (half) * 128, slot_ptr(lds, (op), (buf), (half)), wave, lane)
 
-  // prologue: tile 0 complete + A0,B1 of tile 1 (the halves iteration 0
-  // does not stage itself); full drain once.
+  // prologue: tile 0 complete + A0,B1,A1 of tile 1 (iteration 0 stages
+  // only B0(1) itself, at p1); full drain once.
   STAGE(0, 0, 0, 0);
   STAGE(1, 0, 0, 0);
   STAGE(0, 0, 1, 0);
   STAGE(1, 0, 1, 0);
   STAGE(0, 1, 0, 1);
   STAGE(1, 1, 1, 1);
+  STAGE(0, 1, 1, 1);
   __syncthreads();
 
 #define VM_DRAIN                                                               \
@@ -334,20 +349,20 @@ __global__ __launch_bounds__(512, 1) void mfma_gemm_bf16_256(
     if (tp + 4 >= nk)                                                          \
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");                         \
     else                                                                       \
-      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");                         \
+      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");                         \
   } while (0)
 
   for (int tp = 0; tp < nk; tp += 2) {
     // tile tp from buf0 (reads: 12 / 4 / 8 / 4 ds_read_b128) -----------
-    PHASE(0, 0, 0, 1, 1, acc00, if (tp + 1 < nk) STAGE(0, 1, 1, tp + 1), );
-    PHASE(0, 0, 1, 0, 1, acc01, if (tp + 1 < nk) STAGE(1, 1, 0, tp + 1), );
-    PHASE(0, 1, 1, 1, 0, acc11, if (tp + 2 < nk) STAGE(0, 0, 0, tp + 2), );
-    PHASE(0, 1, 0, 0, 1, acc10, if (tp + 2 < nk) STAGE(1, 0, 1, tp + 2), VM_DRAIN);
+    PHASE(0, 0, 0, 1, 1, acc00, if (tp + 1 < nk) STAGE(1, 1, 0, tp + 1), );
+    PHASE(0, 0, 1, 0, 1, acc01, if (tp + 2 < nk) STAGE(0, 0, 0, tp + 2), );
+    PHASE(0, 1, 1, 1, 0, acc11, if (tp + 2 < nk) STAGE(1, 0, 1, tp + 2), );
+    PHASE(0, 1, 0, 0, 1, acc10, if (tp + 2 < nk) STAGE(0, 0, 1, tp + 2), VM_DRAIN);
     // tile tp+1 from buf1 ----------------------------------------------
-    PHASE(1, 0, 0, 1, 1, acc00, if (tp + 2 < nk) STAGE(0, 0, 1, tp + 2), );
-    PHASE(1, 0, 1, 0, 1, acc01, if (tp + 2 < nk) STAGE(1, 0, 0, tp + 2), );
-    PHASE(1, 1, 1, 1, 0, acc11, if (tp + 3 < nk) STAGE(0, 1, 0, tp + 3), );
-    PHASE(1, 1, 0, 0, 1, acc10, if (tp + 3 < nk) STAGE(1, 1, 1, tp + 3), VM_DRAIN);
+    PHASE(1, 0, 0, 1, 1, acc00, if (tp + 2 < nk) STAGE(1, 0, 0, tp + 2), );
+    PHASE(1, 0, 1, 0, 1, acc01, if (tp + 3 < nk) STAGE(0, 1, 0, tp + 3), );
+    PHASE(1, 1, 1, 1, 0, acc11, if (tp + 3 < nk) STAGE(1, 1, 1, tp + 3), );
+    PHASE(1, 1, 0, 0, 1, acc10, if (tp + 3 < nk) STAGE(0, 1, 1, tp + 3), VM_DRAIN);
   }
 #undef VM_DRAIN
 #undef STAGE
@@ -618,8 +633,12 @@ static int launch_mfma_gemm(const void* A, const void* Bt, void* C, int M,
                             int N, int K) {
   if (M % BM2 == 0 && N % BN2 == 0 && K % (2 * BK2) == 0) {
     dim3 grid(N / BN2, M / BM2);
+    // enable the XCD remap only past the 256 MiB Infinity Cache
+    long ws = 2L * K * (M + N) + 4L * M * N;
+    int swz_on = ws > (256L << 20) ? 1 : 0;
     hipLaunchKernelGGL(mfma_gemm_bf16_256, grid, dim3(512), 0, 0,
-                       (const bf16*)A, (const bf16*)Bt, (float*)C, M, N, K);
+                       (const bf16*)A, (const bf16*)Bt, (float*)C, M, N, K,
+                       swz_on);
     return 0;
   }
   if (M % BM || N % BN || K % BK) return -2;
@@ -649,8 +668,10 @@ int cc_mfma_gemm_bf16_variant(int device, const void* A, const void* Bt,
   if (which == 1) {
     if (M % BM2 || N % BN2 || K % (2 * BK2)) return -2;
     dim3 grid(N / BN2, M / BM2);
+    long ws = 2L * K * (M + N) + 4L * M * N;
     hipLaunchKernelGGL(mfma_gemm_bf16_256, grid, dim3(512), 0, 0,
-                       (const bf16*)A, (const bf16*)Bt, (float*)C, M, N, K);
+                       (const bf16*)A, (const bf16*)Bt, (float*)C, M, N, K,
+                       ws > (256L << 20) ? 1 : 0);
   } else {
     if (M % BM || N % BN || K % BK) return -2;
     dim3 grid(N / BN, M / BM);
