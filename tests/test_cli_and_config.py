@@ -1,0 +1,127 @@
+"""Entry/CLI surface, kubeconfig loading, readiness file, error budget."""
+
+import threading
+
+import pytest
+import yaml
+
+from k8s_cc_manager_amd.core.manager import CCManager, ManagerConfig
+from k8s_cc_manager_amd.core.transition import TransitionEngine
+from k8s_cc_manager_amd.device.mock import MockBackend
+from k8s_cc_manager_amd.k8s.client import ApiError, K8sClient
+from k8s_cc_manager_amd.main import build_parser
+from k8s_cc_manager_amd.utils.readiness import (
+    create_readiness_file,
+    remove_readiness_file,
+)
+
+
+def test_parser_defaults(monkeypatch):
+    monkeypatch.delenv("NODE_NAME", raising=False)
+    monkeypatch.delenv("DEFAULT_CC_MODE", raising=False)
+    args = build_parser().parse_args([])
+    assert args.default_cc_mode == "on"
+    assert args.node_name == ""
+    assert args.device_backend == "auto"
+
+
+def test_parser_env_defaults(monkeypatch):
+    monkeypatch.setenv("NODE_NAME", "worker-3")
+    monkeypatch.setenv("DEFAULT_CC_MODE", "devtools")
+    args = build_parser().parse_args([])
+    assert args.node_name == "worker-3"
+    assert args.default_cc_mode == "devtools"
+
+
+def test_manager_config_from_env(monkeypatch):
+    monkeypatch.setenv("OPERATOR_NAMESPACE", "custom-ns")
+    monkeypatch.setenv("EVICT_OPERATOR_COMPONENTS", "false")
+    monkeypatch.setenv("CORDON_NODE", "false")
+    cfg = ManagerConfig.from_env()
+    assert cfg.operator_namespace == "custom-ns"
+    assert cfg.evict_components is False
+    assert cfg.cordon_node is False
+
+
+def test_kubeconfig_loading(fake_cluster, tmp_path):
+    cluster, url = fake_cluster
+    cluster.add_node("kcnode", labels={"x": "1"})
+    kc = {
+        "apiVersion": "v1",
+        "kind": "Config",
+        "current-context": "test",
+        "contexts": [{"name": "test", "context": {"cluster": "c", "user": "u"}}],
+        "clusters": [{"name": "c", "cluster": {"server": url}}],
+        "users": [{"name": "u", "user": {"token": "dummy-token"}}],
+    }
+    path = tmp_path / "kubeconfig"
+    path.write_text(yaml.safe_dump(kc))
+    client = K8sClient.from_kubeconfig(str(path))
+    node = client.get_node("kcnode")
+    assert node["metadata"]["labels"]["x"] == "1"
+
+
+def test_readiness_file_roundtrip(tmp_path, monkeypatch):
+    f = tmp_path / "deep" / "nested" / ".ready"
+    assert create_readiness_file(str(f))
+    assert f.exists()
+    remove_readiness_file(str(f))
+    assert not f.exists()
+
+
+def test_readiness_file_failure_not_fatal(tmp_path):
+    blocked = tmp_path / "file"
+    blocked.write_text("x")  # parent "dir" is a file -> mkdir fails
+    assert create_readiness_file(str(blocked / "sub" / ".ready")) is False
+
+
+class _FailingClient:
+    """watch always raises; get_node works once (for the resync path)."""
+
+    def __init__(self, inner):
+        self.inner = inner
+        self.watch_calls = 0
+
+    def get_node(self, name):
+        return self.inner.get_node(name)
+
+    def patch_node_labels(self, *a, **kw):
+        return self.inner.patch_node_labels(*a, **kw)
+
+    def set_node_unschedulable(self, *a, **kw):
+        return self.inner.set_node_unschedulable(*a, **kw)
+
+    def list_pods(self, *a, **kw):
+        return self.inner.list_pods(*a, **kw)
+
+    def watch_node(self, *a, **kw):
+        self.watch_calls += 1
+        raise ApiError(500, "injected")
+        yield  # pragma: no cover
+
+    def __getattr__(self, item):  # pragma: no cover
+        return getattr(self.inner, item)
+
+
+def test_watch_error_budget_exhausts(fake_cluster):
+    cluster, url = fake_cluster
+    cluster.add_node("node0")
+    failing = _FailingClient(K8sClient(url))
+    mgr = CCManager(
+        node_name="node0",
+        default_mode="off",
+        host_cc=True,
+        k8s=failing,
+        backend=MockBackend(num_gpus=1),
+        engine=TransitionEngine(),
+        config=ManagerConfig(
+            evict_components=False,
+            cordon_node=False,
+            max_consecutive_errors=3,
+            reconnect_backoff=0.01,
+            readiness_file="/tmp/.cc-test-ready",
+        ),
+    )
+    with pytest.raises(RuntimeError, match="3 times consecutively"):
+        mgr.run()
+    assert failing.watch_calls == 3
