@@ -1,0 +1,31 @@
+#!/usr/bin/env python3
+"""Minimal dispatch stream for PMC collection: N iterations of one GEMM
+variant at one size (rocprofv3 --pmc wraps this)."""
+
+import sys
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+
+import torch  # noqa: E402
+
+from k8s_cc_manager_amd.ops import attest  # noqa: E402
+
+
+def main():
+    n = int(sys.argv[1]) if len(sys.argv) > 1 else 8192
+    which = int(sys.argv[2]) if len(sys.argv) > 2 else 1
+    iters = int(sys.argv[3]) if len(sys.argv) > 3 else 5
+    a = torch.randn(n, n, device="cuda").bfloat16()
+    bt = torch.randn(n, n, device="cuda").bfloat16()
+    c = torch.empty(n, n, device="cuda", dtype=torch.float32)
+    for _ in range(iters):
+        attest.mfma_gemm_bf16_variant(
+            0, a.data_ptr(), bt.data_ptr(), c.data_ptr(), n, n, n, which
+        )
+    torch.cuda.synchronize()
+    print("done")
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,83 @@
+"""Failure-path behavior of the eviction wrapper: unwind semantics the
+reference lacks (it returns early leaving components paused,
+/root/reference/main.py:558-566)."""
+
+import time
+
+from k8s_cc_manager_amd.core.manager import CCManager, ManagerConfig
+from k8s_cc_manager_amd.core.transition import TransitionEngine
+from k8s_cc_manager_amd.device.mock import MockBackend
+from k8s_cc_manager_amd.k8s.client import ApiError, K8sClient
+from k8s_cc_manager_amd.k8s.eviction import COMPONENT_LABELS, PAUSED_VALUE
+
+NODE = "node0"
+
+
+class PatchFaultClient:
+    """Fails patch_node_labels calls that try to PAUSE components."""
+
+    def __init__(self, inner: K8sClient):
+        self.inner = inner
+        self.fail_pause = True
+
+    def patch_node_labels(self, name, labels):
+        if self.fail_pause and any(
+            isinstance(v, str) and PAUSED_VALUE in v for v in labels.values()
+        ):
+            raise ApiError(500, "injected pause failure")
+        return self.inner.patch_node_labels(name, labels)
+
+    def __getattr__(self, item):
+        return getattr(self.inner, item)
+
+
+def _mk(cluster, url, client=None):
+    return CCManager(
+        node_name=NODE,
+        default_mode="on",
+        host_cc=True,
+        k8s=client or K8sClient(url),
+        backend=MockBackend(num_gpus=2),
+        engine=TransitionEngine(),
+        config=ManagerConfig(
+            evict_components=True,
+            cordon_node=True,
+            eviction_timeout=3.0,
+            eviction_poll_interval=0.02,
+        ),
+    )
+
+
+def test_pause_failure_uncordons_and_aborts(fake_cluster):
+    cluster, url = fake_cluster
+    cluster.add_node(NODE, labels={n: "true" for n in COMPONENT_LABELS})
+    client = PatchFaultClient(K8sClient(url))
+    mgr = _mk(cluster, url, client)
+    assert mgr.apply_mode("on") is False
+    # no device work happened
+    assert all(m == "off" for m in mgr.backend.modes().values())
+    # node not left cordoned
+    assert not cluster.node_unschedulable(NODE)
+    # component labels untouched (pause never landed)
+    for name in COMPONENT_LABELS:
+        assert cluster.node_labels(NODE)[name] == "true"
+
+
+def test_drain_timeout_is_nonfatal(fake_cluster):
+    """Pods that never drain: eviction logs and proceeds (reference
+    envelope g_o_e.py:205-207) — transition still runs and components
+    are restored afterwards."""
+    cluster, url = fake_cluster
+    cluster.add_node(NODE, labels={n: "true" for n in COMPONENT_LABELS})
+    # a stuck pod the operator simulator will NOT delete (unknown app)
+    cluster.add_pod(cluster.operator_namespace, "stuck", NODE, app="amd-gpu-device-plugin")
+    # freeze the operator so the pod never drains
+    cluster._operator_tick = 999
+    mgr = _mk(cluster, url)
+    mgr.config.eviction_timeout = 0.3
+    t0 = time.monotonic()
+    assert mgr.apply_mode("on") is True
+    assert time.monotonic() - t0 < 5
+    assert all(m == "on" for m in mgr.backend.modes().values())
+    for name in COMPONENT_LABELS:
+        assert cluster.node_labels(NODE)[name] == "true"  # restored
