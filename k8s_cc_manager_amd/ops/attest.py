@@ -214,6 +214,10 @@ def attest_device_by_bdf(device) -> None:
     gemm_dim = int(os.environ.get("CC_ATTEST_GEMM_DIM", "1024"))
     idx = device_index_for_bdf(device.bdf)
     attest_device(idx, gemm_dim=gemm_dim)
+    if os.environ.get("CC_ATTEST_DEEP", "0") == "1":
+        from .deep_attest import deep_attest_device
+
+        deep_attest_device(idx, gemm_dim=gemm_dim)
 
 
 def mfma_gemm_bf16(device_index: int, a_ptr: int, bt_ptr: int, c_ptr: int,

@@ -1,0 +1,37 @@
+"""CPU-side tests of the deep-attestation CSV parsing."""
+
+import pytest
+
+from k8s_cc_manager_amd.ops.attest import AttestationError
+from k8s_cc_manager_amd.ops.deep_attest import MFMA_COUNTER, _sum_counter_csv
+
+
+def test_sum_counter_csv(tmp_path):
+    d = tmp_path / "out"
+    d.mkdir()
+    (d / "probe_counter_collection.csv").write_text(
+        "Dispatch_Id,Counter_Name,Counter_Value\n"
+        f"1,{MFMA_COUNTER},1000\n"
+        f"2,{MFMA_COUNTER},2500\n"
+        "3,OTHER_COUNTER,999\n"
+    )
+    assert _sum_counter_csv(d, MFMA_COUNTER) == 3500.0
+
+
+def test_sum_counter_csv_missing_counter(tmp_path):
+    d = tmp_path / "out"
+    d.mkdir()
+    (d / "x_counter_collection.csv").write_text(
+        "Dispatch_Id,Counter_Name,Counter_Value\n1,OTHER,1\n"
+    )
+    with pytest.raises(AttestationError):
+        _sum_counter_csv(d, MFMA_COUNTER)
+
+
+def test_sum_counter_csv_nested_dirs(tmp_path):
+    d = tmp_path / "out" / "pid123"
+    d.mkdir(parents=True)
+    (d / "probe_counter_collection.csv").write_text(
+        f"Counter_Name,Counter_Value\n{MFMA_COUNTER},7\n"
+    )
+    assert _sum_counter_csv(tmp_path / "out", MFMA_COUNTER) == 7.0

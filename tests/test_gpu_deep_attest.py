@@ -1,0 +1,17 @@
+"""Deep attestation (rocprof counter readback) on a real MI355X."""
+
+import pytest
+
+pytestmark = [pytest.mark.gpu]
+
+
+def test_deep_attest_counts_mfma_cycles():
+    from k8s_cc_manager_amd.ops.deep_attest import (
+        deep_attest_device,
+        rocprof_available,
+    )
+
+    if not rocprof_available():
+        pytest.skip("rocprofv3 not installed")
+    cycles = deep_attest_device(0, gemm_dim=512, timeout=240)
+    assert cycles > 1e6, f"implausibly few MFMA busy cycles: {cycles}"
