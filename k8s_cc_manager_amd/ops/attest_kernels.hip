@@ -95,8 +95,14 @@ __global__ void fill_f32_iota(float* __restrict__ out, long n, float scale) {
 constexpr int TILE_B = BM * BK * 2;  // 16 KiB per operand tile
 
 __device__ __forceinline__ int swz(int byte_off) {
-  // st_16x32 swizzle: XOR byte-bit-5 with bit-9 inside each 1 KiB subtile
-  return byte_off ^ (((byte_off >> 9) & 1) << 5);
+  // 3-bit XOR swizzle: inject row bits 1-3 (byte bits 8-10 of the
+  // 128-B-row image) into bank bits 2-4 (byte bits 4-6). With the
+  // fragment pattern row=(lane&15|31), khalf=(lane>>4|5)*16B this puts
+  // the 16 lanes of every ds_read_b128 lane group on 16 distinct
+  // 4-bank quads (enumerated for all four groups) — the 1-bit st_16x32
+  // form left rows 0-3 on two quads (2-way conflict, ~9% of cycles in
+  // the PMC profile, profiles/v256_8192_pmc.txt).
+  return byte_off ^ (((byte_off >> 8) & 7) << 4);
 }
 
 __device__ __forceinline__ void stage_tile_glds(
@@ -393,6 +399,132 @@ __global__ __launch_bounds__(512, 1) void mfma_gemm_bf16_256(
 }
 
 // ---------------------------------------------------------------------------
+// Same 8-phase 256x256 structure on v_mfma_f32_32x32x16_bf16 (the
+// higher-ceiling MFMA shape: 2382 vs 2075 TF/s µbench). Per quadrant a
+// wave computes 2 m-tiles of 32x32 over 4 k-steps of 16 (8 MFMA/phase).
+// Fragment maps: A row=l&31, k=(l>>5)*8+j; Bt col=l&31 likewise;
+// C/D col=lane&31, row=(reg&3)+8*(reg>>2)+4*(lane>>5).
+// ---------------------------------------------------------------------------
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+
+#define PHASE32(buf, mh, nh, ACC, PREFETCH_STMT, DRAIN, LOAD_A, LOAD_B)        \
+  do {                                                                         \
+    if (LOAD_A) {                                                              \
+      char* Ah = slot_ptr(lds, 0, (buf), (mh));                                \
+      _Pragma("unroll") for (int ks = 0; ks < 4; ++ks)                         \
+          _Pragma("unroll") for (int t = 0; t < 2; ++t) {                      \
+        int lg = (wave_mq + t * 32 + lane31) * 128 + ks * 32 + khalf32_b;      \
+        a2[t][ks] = *(const bf16x8*)(Ah + swz(lg));                            \
+      }                                                                        \
+    }                                                                          \
+    if (LOAD_B) {                                                              \
+      char* Bh = slot_ptr(lds, 1, (buf), (nh));                                \
+      _Pragma("unroll") for (int ks = 0; ks < 4; ++ks) {                       \
+        int lg = (wave_nq + lane31) * 128 + ks * 32 + khalf32_b;               \
+        b2[ks] = *(const bf16x8*)(Bh + swz(lg));                               \
+      }                                                                        \
+    }                                                                          \
+    PREFETCH_STMT;                                                             \
+    DRAIN;                                                                     \
+    __builtin_amdgcn_s_barrier();                                              \
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                         \
+    __builtin_amdgcn_s_setprio(1);                                             \
+    _Pragma("unroll") for (int ks = 0; ks < 4; ++ks)                           \
+        _Pragma("unroll") for (int t = 0; t < 2; ++t)                          \
+            ACC[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(                  \
+                a2[t][ks], b2[ks], ACC[t], 0, 0, 0);                           \
+    __builtin_amdgcn_s_setprio(0);                                             \
+    __builtin_amdgcn_s_barrier();                                              \
+  } while (0)
+
+__global__ __launch_bounds__(512, 1) void mfma_gemm_bf16_256w(
+    const bf16* __restrict__ A, const bf16* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K, int xcd_swizzle) {
+  __shared__ char lds[8 * HALF_B];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wave_mq = (wave >> 2) * 64;
+  const int wave_nq = (wave & 3) * 32;
+  const int lane31 = lane & 31;
+  const int khalf32_b = (lane >> 5) * 16;
+  int wg = blockIdx.y * gridDim.x + blockIdx.x;
+  if (xcd_swizzle) {
+    int nwg = gridDim.x * gridDim.y;
+    int q = nwg >> 3, r = nwg & 7;
+    int xcd = wg & 7, o = wg >> 3;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + o;
+  }
+  const int block_m = (wg / gridDim.x) * BM2;
+  const int block_n = (wg % gridDim.x) * BN2;
+
+  const char* gA = (const char*)(A + (long)block_m * K);
+  const char* gB = (const char*)(Bt + (long)block_n * K);
+  const long row_b = (long)K * 2;
+  const int nk = K / BK2;
+
+  f32x16 acc00[2] = {}, acc01[2] = {}, acc10[2] = {}, acc11[2] = {};
+  bf16x8 a2[2][4], b2[4];
+
+#define STAGE(op, buf, half, tile)                                             \
+  stage_half_glds((op) == 0 ? gA : gB, row_b, (long)(tile) * BK2 * 2,          \
+                  (half) * 128, slot_ptr(lds, (op), (buf), (half)), wave, lane)
+
+  STAGE(0, 0, 0, 0);
+  STAGE(1, 0, 0, 0);
+  STAGE(0, 0, 1, 0);
+  STAGE(1, 0, 1, 0);
+  STAGE(0, 1, 0, 1);
+  STAGE(1, 1, 1, 1);
+  STAGE(0, 1, 1, 1);
+  __syncthreads();
+
+#define VM_DRAIN                                                               \
+  do {                                                                         \
+    if (tp + 4 >= nk)                                                          \
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");                         \
+    else                                                                       \
+      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");                         \
+  } while (0)
+
+  for (int tp = 0; tp < nk; tp += 2) {
+    PHASE32(0, 0, 0, acc00, if (tp + 1 < nk) STAGE(1, 1, 0, tp + 1), , 1, 1);
+    PHASE32(0, 0, 1, acc01, if (tp + 2 < nk) STAGE(0, 0, 0, tp + 2), , 0, 1);
+    PHASE32(0, 1, 1, acc11, if (tp + 2 < nk) STAGE(1, 0, 1, tp + 2), , 1, 0);
+    PHASE32(0, 1, 0, acc10, if (tp + 2 < nk) STAGE(0, 0, 1, tp + 2), VM_DRAIN, 0, 1);
+    PHASE32(1, 0, 0, acc00, if (tp + 2 < nk) STAGE(1, 0, 0, tp + 2), , 1, 1);
+    PHASE32(1, 0, 1, acc01, if (tp + 3 < nk) STAGE(0, 1, 0, tp + 3), , 0, 1);
+    PHASE32(1, 1, 1, acc11, if (tp + 3 < nk) STAGE(1, 1, 1, tp + 3), , 1, 0);
+    PHASE32(1, 1, 0, acc10, if (tp + 3 < nk) STAGE(0, 1, 1, tp + 3), VM_DRAIN, 0, 1);
+  }
+#undef VM_DRAIN
+#undef STAGE
+
+  // epilogue: C/D col=lane&31, row=(reg&3)+8*(reg>>2)+4*(lane>>5)
+  const int c_col32 = lane & 31;
+  const int c_rowhi = (lane >> 5) * 4;
+#pragma unroll
+  for (int mh = 0; mh < 2; ++mh) {
+#pragma unroll
+    for (int nh = 0; nh < 2; ++nh) {
+      f32x16* accq = mh == 0 ? (nh == 0 ? acc00 : acc01)
+                             : (nh == 0 ? acc10 : acc11);
+#pragma unroll
+      for (int t = 0; t < 2; ++t) {
+#pragma unroll
+        for (int reg = 0; reg < 16; ++reg) {
+          int row = block_m + mh * 128 + wave_mq + t * 32 + (reg & 3) +
+                    8 * (reg >> 2) + c_rowhi;
+          int col = block_n + nh * 128 + wave_nq + c_col32;
+          C[(long)row * N + col] = accq[t][reg];
+        }
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // VALU fp32 reference GEMM (independent ground truth; deliberately does
 // NOT share tiling or fragment code with the MFMA path).
 // One thread per C element, fp32 FMA chain over K.
@@ -665,13 +797,19 @@ int cc_mfma_gemm_bf16(int device, const void* A, const void* Bt, void* C,
 int cc_mfma_gemm_bf16_variant(int device, const void* A, const void* Bt,
                               void* C, int M, int N, int K, int which) {
   if (hipSetDevice(device) != hipSuccess) return -3;
-  if (which == 1) {
+  if (which == 1 || which == 2) {
     if (M % BM2 || N % BN2 || K % (2 * BK2)) return -2;
     dim3 grid(N / BN2, M / BM2);
     long ws = 2L * K * (M + N) + 4L * M * N;
-    hipLaunchKernelGGL(mfma_gemm_bf16_256, grid, dim3(512), 0, 0,
-                       (const bf16*)A, (const bf16*)Bt, (float*)C, M, N, K,
-                       ws > (256L << 20) ? 1 : 0);
+    int swz_on = ws > (256L << 20) ? 1 : 0;
+    if (which == 2)
+      hipLaunchKernelGGL(mfma_gemm_bf16_256w, grid, dim3(512), 0, 0,
+                         (const bf16*)A, (const bf16*)Bt, (float*)C, M, N, K,
+                         swz_on);
+    else
+      hipLaunchKernelGGL(mfma_gemm_bf16_256, grid, dim3(512), 0, 0,
+                         (const bf16*)A, (const bf16*)Bt, (float*)C, M, N, K,
+                         swz_on);
   } else {
     if (M % BM || N % BN || K % BK) return -2;
     dim3 grid(N / BN, M / BM);

