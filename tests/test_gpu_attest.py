@@ -60,8 +60,9 @@ def test_mfma_gemm_integer_exact(attest):
     assert torch.equal(c, ref)
 
 
+@pytest.mark.parametrize("which", [1, 2])
 @pytest.mark.parametrize("m,n,k", [(256, 256, 128), (512, 256, 384), (256, 512, 1024)])
-def test_mfma_gemm_256_template_matches_torch(attest, m, n, k):
+def test_mfma_gemm_256_template_matches_torch(attest, m, n, k, which):
     """The deep-pipelined 256x256 8-phase variant — race-sensitive, so
     run it several times per shape (sync-structure screening)."""
     torch.manual_seed(m * 7 + n * 3 + k)
@@ -72,11 +73,11 @@ def test_mfma_gemm_256_template_matches_torch(attest, m, n, k):
     for trial in range(5):
         c = torch.full((m, n), float("nan"), device="cuda", dtype=torch.float32)
         attest.mfma_gemm_bf16_variant(
-            0, a.data_ptr(), bt.data_ptr(), c.data_ptr(), m, n, k, 1
+            0, a.data_ptr(), bt.data_ptr(), c.data_ptr(), m, n, k, which
         )
         torch.cuda.synchronize()
         err = (c - ref).abs().max().item()
-        assert err <= 2e-3 * max(scale, 1.0), f"trial {trial}: err={err}"
+        assert err <= 2e-3 * max(scale, 1.0), f"trial {trial} which={which}: err={err}"
 
 
 def test_mfma_gemm_256_integer_exact_race_screen(attest):
