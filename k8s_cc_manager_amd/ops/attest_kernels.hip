@@ -22,8 +22,8 @@
 // fp32 of C/D; A lane mapping row=l&15, k=(l>>4)*8+j; B (we require the
 // second operand TRANSPOSED, i.e. Bt[N][K], so its lane mapping is the
 // same as A's with col=l&15); C/D mapping col=l&15, row=(l>>4)*4+r.
-// LDS tile images carry the st_16x32 XOR swizzle (byte-bit-5 ^= bit-9)
-// so ds_read_b128 fragment reads are bank-conflict-free while the
+// LDS tile images carry a 3-bit XOR bank swizzle (see swz()) so
+// ds_read_b128 fragment reads are bank-conflict-free while the
 // global->LDS DMA stays lane-linear (swizzle rides the source address).
 
 #include <hip/hip_runtime.h>
@@ -72,12 +72,6 @@ __global__ void fill_bf16_lcg(bf16* __restrict__ out, long n, uint32_t seed) {
     int v = (int)((x >> 13) & 3u) - 2;  // {-2,-1,0,1}
     out[i] = (bf16)(float)v;
   }
-}
-
-__global__ void fill_f32_iota(float* __restrict__ out, long n, float scale) {
-  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  long stride = (long)gridDim.x * blockDim.x;
-  for (; i < n; i += stride) out[i] = scale * (float)(i & 1023);
 }
 
 // ---------------------------------------------------------------------------
@@ -254,13 +248,6 @@ __device__ __forceinline__ void stage_half_glds(
         (__attribute__((address_space(3))) void*)(lds_half + base), 16, 0, 0);
   }
 }
-
-struct Gemm256Args {
-  const char* gA;
-  const char* gB;
-  long row_b;
-  int nk;
-};
 
 // one phase: ds-read only the fragments this quadrant does NOT already
 // hold (adjacent phases share an A-half or a B-half of the same K-tile,
