@@ -29,12 +29,11 @@ from typing import Callable, Dict, List, Optional, Sequence
 
 from ..device.contract import (
     CCDevice,
-    CCDeviceError,
     FABRIC_OFF,
     FABRIC_ON,
     ModeVerifyError,
 )
-from ..parallel.executor import DeviceExecutor, PerDeviceError
+from ..parallel.executor import DeviceExecutor
 from ..utils.timing import PhaseTimer
 
 logger = logging.getLogger(__name__)
@@ -146,7 +145,7 @@ class TransitionEngine:
                 ),
             )
             timer.stop()
-        except (PerDeviceError, CCDeviceError, Exception) as e:
+        except Exception as e:
             timer.stop()
             report.seconds = timer.total()
             report.phases = timer.as_dict()
@@ -223,7 +222,7 @@ class TransitionEngine:
                 ),
             )
             timer.stop()
-        except (PerDeviceError, CCDeviceError, Exception) as e:
+        except Exception as e:
             timer.stop()
             report.seconds = timer.total()
             report.phases = timer.as_dict()

@@ -249,6 +249,19 @@ __device__ __forceinline__ void stage_half_glds(
   }
 }
 
+// Experiment toggles (A/B builds; default off):
+//   CC_EXP_PREFETCH_FIRST — issue the half-tile DMA before the phase's
+//     ds_reads (DMA in flight during LDS reads);
+//   CC_EXP_LGKM — partial s_waitcnt lgkmcnt(4) before the barrier so
+//     waves arrive with most reads drained (the template's optional
+//     pre-wait on 12-read phases).
+#ifndef CC_EXP_PREFETCH_FIRST
+#define CC_EXP_PREFETCH_FIRST 0
+#endif
+#ifndef CC_EXP_LGKM
+#define CC_EXP_LGKM 0
+#endif
+
 // one phase: ds-read only the fragments this quadrant does NOT already
 // hold (adjacent phases share an A-half or a B-half of the same K-tile,
 // so those fragments persist in VGPRs across the phase barrier: phases
@@ -256,6 +269,9 @@ __device__ __forceinline__ void stage_half_glds(
 // raw barrier, lgkmcnt(0), 16 MFMA at prio 1, raw barrier.
 #define PHASE(buf, mh, nh, LOAD_A, LOAD_B, ACC, PREFETCH_STMT, DRAIN)          \
   do {                                                                         \
+    if (CC_EXP_PREFETCH_FIRST) {                                               \
+      PREFETCH_STMT;                                                           \
+    }                                                                          \
     if (LOAD_A) {                                                              \
       char* Ah = slot_ptr(lds, 0, (buf), (mh));                                \
       _Pragma("unroll") for (int ks = 0; ks < 2; ++ks)                         \
@@ -272,8 +288,12 @@ __device__ __forceinline__ void stage_half_glds(
         bf[j][ks] = *(const bf16x8*)(Bh + swz(lg));                            \
       }                                                                        \
     }                                                                          \
-    PREFETCH_STMT;                                                             \
+    if (!CC_EXP_PREFETCH_FIRST) {                                              \
+      PREFETCH_STMT;                                                           \
+    }                                                                          \
     DRAIN;                                                                     \
+    if (CC_EXP_LGKM && ((LOAD_A) + (LOAD_B)) > 1)                              \
+      asm volatile("s_waitcnt lgkmcnt(4)" ::: "memory");                       \
     __builtin_amdgcn_s_barrier();                                              \
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                         \
     __builtin_amdgcn_s_setprio(1);                                             \
