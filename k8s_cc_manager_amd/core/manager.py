@@ -114,6 +114,20 @@ class CCManager:
     def _set_state(self, state: str) -> None:
         eviction.set_cc_state_label(self.k8s, self.node_name, state)
 
+    def publish_capability_label(self) -> None:
+        """Advertise whether this node can do GPU-CC at all:
+        ``amd.com/gpu.cc.capable`` = host TEE support AND >=1 CC-capable
+        GPU. Informational (the operator's node-labeller may also set
+        it); best-effort."""
+        try:
+            capable = self.host_cc and bool(self.backend.get_cc_capable_gpus())
+            self.k8s.patch_node_labels(
+                self.node_name,
+                {"amd.com/gpu.cc.capable": "true" if capable else "false"},
+            )
+        except Exception as e:
+            logger.warning("could not publish capability label: %s", e)
+
     # ------------------------------------------------------------------
     # mode application (reference dispatcher: main.py:214-263)
     # ------------------------------------------------------------------
@@ -259,6 +273,7 @@ class CCManager:
     def run(self) -> None:
         self.read_mode_label()
         self.apply_mode(self.with_default(self.current_label))
+        self.publish_capability_label()
         create_readiness_file(self.config.readiness_file)
 
         last_applied = self.current_label

@@ -58,6 +58,12 @@ def build_parser() -> argparse.ArgumentParser:
         default=int(os.environ.get("CC_METRICS_PORT", "0")),
         help="serve Prometheus metrics on this port (0 = off)",
     )
+    parser.add_argument(
+        "--once",
+        action="store_true",
+        help="apply the current desired mode once and exit (no watch loop); "
+        "exit code reflects the transition result",
+    )
     parser.add_argument("--debug", action="store_true", help="debug logging")
     return parser
 
@@ -100,6 +106,11 @@ def main(argv=None) -> int:
         config=ManagerConfig.from_env(),
     )
     try:
+        if args.once:
+            label = manager.read_mode_label()
+            ok = manager.apply_mode(manager.with_default(label))
+            manager.publish_capability_label()
+            return 0 if ok else 1
         manager.run()
         return 0
     except KeyboardInterrupt:
