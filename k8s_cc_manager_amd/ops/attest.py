@@ -180,7 +180,26 @@ def attest_device(device_index: int, gemm_dim: int = 1024) -> AttestReport:
         rep.peers_accessible,
         rep.peer_count,
     )
+    _append_attest_log(rep)
     return rep
+
+
+def _append_attest_log(rep: AttestReport) -> None:
+    """Append the report as JSONL when CC_ATTEST_LOG is set (audit
+    trail of what each readiness decision was based on)."""
+    path = os.environ.get("CC_ATTEST_LOG")
+    if not path:
+        return
+    import dataclasses
+    import json
+    import time
+
+    try:
+        entry = {"ts": time.time(), **dataclasses.asdict(rep)}
+        with open(path, "a") as f:
+            f.write(json.dumps(entry) + "\n")
+    except OSError as e:  # pragma: no cover
+        logger.debug("attest log write failed: %s", e)
 
 
 _BDF_RE = re.compile(
