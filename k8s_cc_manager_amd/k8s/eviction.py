@@ -131,7 +131,12 @@ def evict_components(
         for name, v in current_labels.items()
         if v and v != "false" and name in COMPONENT_APP_LABELS
     }
+    # Adaptive poll: start fast (fast drains finish in ms), back off
+    # exponentially toward ``poll_interval`` so slow drains do not
+    # hammer the API server (the reference polls at a fixed 2 s,
+    # g_o_e.py:200).
     deadline = time.monotonic() + timeout
+    delay = min(0.002, poll_interval)
     while pending and time.monotonic() < deadline:
         for app in sorted(pending):
             try:
@@ -150,7 +155,8 @@ def evict_components(
             else:
                 logger.debug("%s: %d pod(s) remaining", app, n)
         if pending:
-            time.sleep(poll_interval)
+            time.sleep(delay)
+            delay = min(delay * 2, poll_interval)
 
     if pending:
         logger.warning("drain deadline passed with pods remaining: %s", sorted(pending))
