@@ -134,6 +134,8 @@ def _load() -> ctypes.CDLL:
     lib.cc_mfma_gemm_bf16_variant.argtypes = lib.cc_mfma_gemm_bf16.argtypes + [
         ctypes.c_int
     ]
+    lib.cc_mfma_gemm_fp8.restype = ctypes.c_int
+    lib.cc_mfma_gemm_fp8.argtypes = lib.cc_mfma_gemm_bf16.argtypes
     _lib = lib
     return lib
 
@@ -253,6 +255,15 @@ def ref_gemm_f32(device_index: int, a_ptr: int, bt_ptr: int, c_ptr: int,
     rc = _load().cc_ref_gemm_f32(device_index, a_ptr, bt_ptr, c_ptr, m, n, k)
     if rc != 0:
         raise AttestationError(f"ref_gemm_f32 rc={rc}")
+
+
+def mfma_gemm_fp8(device_index: int, a_ptr: int, bt_ptr: int, c_ptr: int,
+                  m: int, n: int, k: int) -> None:
+    """MX-scaled fp8 e4m3 GEMM (unit scales): C = A @ Bt^T, fp32 out.
+    M,N multiples of 256, K of 256."""
+    rc = _load().cc_mfma_gemm_fp8(device_index, a_ptr, bt_ptr, c_ptr, m, n, k)
+    if rc != 0:
+        raise AttestationError(f"mfma_gemm_fp8 rc={rc}")
 
 
 def mfma_gemm_bf16_variant(device_index: int, a_ptr: int, bt_ptr: int,

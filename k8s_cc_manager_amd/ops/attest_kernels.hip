@@ -532,6 +532,149 @@ __global__ __launch_bounds__(512, 1) void mfma_gemm_bf16_256w(
 }
 
 // ---------------------------------------------------------------------------
+// MX-scaled fp8 (OCP e4m3) GEMM on the same 8-phase 256x256 structure:
+// v_mfma_scale_f32_32x32x64_f8f6f4 with unit e8m0 scales — the only
+// path to the ~5 PF fp8 rate on gfx950 (non-scaled fp8 MFMA runs at
+// the bf16 rate). BK = 128 fp8 elements = the same 128-B row image and
+// 16 KiB half-tiles as the bf16 template, so the phase schedule,
+// prefetch map, vmcnt drains and bank swizzle carry over unchanged.
+// Per phase a wave runs 4 MFMA (2 m-tiles x 2 k-steps of 64).
+// Fragment map: A row=l&31, k=(l>>5)*32+j (32 consecutive fp8 = two
+// ds_read_b128); Bt col=l&31 likewise; C/D layout is shape-determined
+// (same as 32x32 bf16).
+// ---------------------------------------------------------------------------
+typedef __attribute__((ext_vector_type(8))) int v8i;
+typedef __attribute__((ext_vector_type(4))) int v4i;
+
+constexpr int BK8 = 128;                 // fp8 elements per K-tile
+constexpr unsigned SCALE_ONE = 0x7F7F7F7Fu;  // e8m0 bias-127 = 2^0 per byte
+
+__device__ __forceinline__ v8i load_frag32(const char* p) {
+  v4i lo = *(const v4i*)(p);
+  v4i hi = *(const v4i*)(p + 16);
+  v8i out;
+  out[0] = lo[0]; out[1] = lo[1]; out[2] = lo[2]; out[3] = lo[3];
+  out[4] = hi[0]; out[5] = hi[1]; out[6] = hi[2]; out[7] = hi[3];
+  return out;
+}
+
+#define PHASE8(buf, mh, nh, ACC, PREFETCH_STMT, DRAIN, LOAD_A, LOAD_B)         \
+  do {                                                                         \
+    if (LOAD_A) {                                                              \
+      char* Ah = slot_ptr(lds, 0, (buf), (mh));                                \
+      _Pragma("unroll") for (int ks = 0; ks < 2; ++ks)                         \
+          _Pragma("unroll") for (int t = 0; t < 2; ++t) {                      \
+        int lg = (wave_mq + t * 32 + lane31) * 128 + ks * 64 + kq_b;           \
+        a8[t][ks] = load_frag32(Ah + swz(lg));                                 \
+      }                                                                        \
+    }                                                                          \
+    if (LOAD_B) {                                                              \
+      char* Bh = slot_ptr(lds, 1, (buf), (nh));                                \
+      _Pragma("unroll") for (int ks = 0; ks < 2; ++ks) {                       \
+        int lg = (wave_nq + lane31) * 128 + ks * 64 + kq_b;                    \
+        b8[ks] = load_frag32(Bh + swz(lg));                                    \
+      }                                                                        \
+    }                                                                          \
+    PREFETCH_STMT;                                                             \
+    DRAIN;                                                                     \
+    __builtin_amdgcn_s_barrier();                                              \
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                         \
+    __builtin_amdgcn_s_setprio(1);                                             \
+    _Pragma("unroll") for (int ks = 0; ks < 2; ++ks)                           \
+        _Pragma("unroll") for (int t = 0; t < 2; ++t)                          \
+            ACC[t] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(          \
+                a8[t][ks], b8[ks], ACC[t], 0, 0, 0, SCALE_ONE, 0, SCALE_ONE);  \
+    __builtin_amdgcn_s_setprio(0);                                             \
+    __builtin_amdgcn_s_barrier();                                              \
+  } while (0)
+
+__global__ __launch_bounds__(512, 1) void mfma_gemm_fp8_256(
+    const char* __restrict__ A, const char* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K, int xcd_swizzle) {
+  __shared__ char lds[8 * HALF_B];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wave_mq = (wave >> 2) * 64;
+  const int wave_nq = (wave & 3) * 32;
+  const int lane31 = lane & 31;
+  const int kq_b = (lane >> 5) * 32;  // fragment k byte offset (32 fp8)
+  int wg = blockIdx.y * gridDim.x + blockIdx.x;
+  if (xcd_swizzle) {
+    int nwg = gridDim.x * gridDim.y;
+    int q = nwg >> 3, r = nwg & 7;
+    int xcd = wg & 7, o = wg >> 3;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + o;
+  }
+  const int block_m = (wg / gridDim.x) * BM2;
+  const int block_n = (wg % gridDim.x) * BN2;
+
+  const char* gA = A + (long)block_m * K;
+  const char* gB = Bt + (long)block_n * K;
+  const long row_b = (long)K;  // 1 byte per element
+  const int nk = K / BK8;
+
+  f32x16 acc00[2] = {}, acc01[2] = {}, acc10[2] = {}, acc11[2] = {};
+  v8i a8[2][2], b8[2];
+
+#define STAGE(op, buf, half, tile)                                             \
+  stage_half_glds((op) == 0 ? gA : gB, row_b, (long)(tile) * BK8, (half) * 128,\
+                  slot_ptr(lds, (op), (buf), (half)), wave, lane)
+
+  STAGE(0, 0, 0, 0);
+  STAGE(1, 0, 0, 0);
+  STAGE(0, 0, 1, 0);
+  STAGE(1, 0, 1, 0);
+  STAGE(0, 1, 0, 1);
+  STAGE(1, 1, 1, 1);
+  STAGE(0, 1, 1, 1);
+  __syncthreads();
+
+#define VM_DRAIN                                                               \
+  do {                                                                         \
+    if (tp + 4 >= nk)                                                          \
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");                         \
+    else                                                                       \
+      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");                         \
+  } while (0)
+
+  for (int tp = 0; tp < nk; tp += 2) {
+    PHASE8(0, 0, 0, acc00, if (tp + 1 < nk) STAGE(1, 1, 0, tp + 1), , 1, 1);
+    PHASE8(0, 0, 1, acc01, if (tp + 2 < nk) STAGE(0, 0, 0, tp + 2), , 0, 1);
+    PHASE8(0, 1, 1, acc11, if (tp + 2 < nk) STAGE(1, 0, 1, tp + 2), , 1, 0);
+    PHASE8(0, 1, 0, acc10, if (tp + 2 < nk) STAGE(0, 0, 1, tp + 2), VM_DRAIN, 0, 1);
+    PHASE8(1, 0, 0, acc00, if (tp + 2 < nk) STAGE(1, 0, 0, tp + 2), , 1, 1);
+    PHASE8(1, 0, 1, acc01, if (tp + 3 < nk) STAGE(0, 1, 0, tp + 3), , 0, 1);
+    PHASE8(1, 1, 1, acc11, if (tp + 3 < nk) STAGE(1, 1, 1, tp + 3), , 1, 0);
+    PHASE8(1, 1, 0, acc10, if (tp + 3 < nk) STAGE(0, 1, 1, tp + 3), VM_DRAIN, 0, 1);
+  }
+#undef VM_DRAIN
+#undef STAGE
+
+  const int c_col32 = lane & 31;
+  const int c_rowhi = (lane >> 5) * 4;
+#pragma unroll
+  for (int mh = 0; mh < 2; ++mh) {
+#pragma unroll
+    for (int nh = 0; nh < 2; ++nh) {
+      f32x16* accq = mh == 0 ? (nh == 0 ? acc00 : acc01)
+                             : (nh == 0 ? acc10 : acc11);
+#pragma unroll
+      for (int t = 0; t < 2; ++t) {
+#pragma unroll
+        for (int reg = 0; reg < 16; ++reg) {
+          int row = block_m + mh * 128 + wave_mq + t * 32 + (reg & 3) +
+                    8 * (reg >> 2) + c_rowhi;
+          int col = block_n + nh * 128 + wave_nq + c_col32;
+          C[(long)row * N + col] = accq[t][reg];
+        }
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // VALU fp32 reference GEMM (independent ground truth; deliberately does
 // NOT share tiling or fragment code with the MFMA path).
 // One thread per C element, fp32 FMA chain over K.
@@ -823,6 +966,20 @@ int cc_mfma_gemm_bf16_variant(int device, const void* A, const void* Bt,
     hipLaunchKernelGGL(mfma_gemm_bf16, grid, dim3(256), 0, 0, (const bf16*)A,
                        (const bf16*)Bt, (float*)C, M, N, K);
   }
+  return (int)hipDeviceSynchronize();
+}
+
+// MX-scaled fp8 (e4m3) GEMM: C[M,N] = A[M,K] @ Bt[N,K]^T, fp8 inputs,
+// fp32 out. M,N multiples of 256; K multiple of 256.
+int cc_mfma_gemm_fp8(int device, const void* A, const void* Bt, void* C,
+                     int M, int N, int K) {
+  if (M % BM2 || N % BN2 || K % (2 * BK8)) return -2;
+  if (hipSetDevice(device) != hipSuccess) return -3;
+  dim3 grid(N / BN2, M / BM2);
+  long ws = (long)K * (M + N) + 4L * M * N;
+  hipLaunchKernelGGL(mfma_gemm_fp8_256, grid, dim3(512), 0, 0, (const char*)A,
+                     (const char*)Bt, (float*)C, M, N, K,
+                     ws > (256L << 20) ? 1 : 0);
   return (int)hipDeviceSynchronize();
 }
 

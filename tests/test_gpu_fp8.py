@@ -1,0 +1,52 @@
+"""fp8 (MX-scaled e4m3, unit scales) GEMM on a real MI355X."""
+
+import pytest
+
+torch = pytest.importorskip("torch")
+
+pytestmark = [
+    pytest.mark.gpu,
+    pytest.mark.skipif(not torch.cuda.is_available(), reason="needs an MI355X"),
+]
+
+
+@pytest.fixture(scope="module")
+def attest():
+    from k8s_cc_manager_amd.ops import attest as a
+
+    assert a.probe_available()
+    return a
+
+
+def _fp8(t):
+    return t.to(torch.float8_e4m3fn)
+
+
+def test_fp8_gemm_integer_exact(attest):
+    """Small integers are exact in e4m3; fp32 accumulation on both
+    sides -> bitwise agreement with the torch fp32 reference."""
+    m = n = 256
+    k = 512
+    torch.manual_seed(7)
+    a = _fp8(torch.randint(-2, 2, (m, k), device="cuda").float())
+    bt = _fp8(torch.randint(-2, 2, (n, k), device="cuda").float())
+    ref = a.float() @ bt.float().t()
+    for trial in range(5):
+        c = torch.full((m, n), float("nan"), device="cuda", dtype=torch.float32)
+        attest.mfma_gemm_fp8(0, a.data_ptr(), bt.data_ptr(), c.data_ptr(), m, n, k)
+        torch.cuda.synchronize()
+        assert torch.equal(c, ref), f"trial {trial}: max diff {(c-ref).abs().max()}"
+
+
+@pytest.mark.parametrize("m,n,k", [(256, 512, 256), (512, 256, 1024)])
+def test_fp8_gemm_random_matches_reference(attest, m, n, k):
+    torch.manual_seed(m + n + k)
+    a = _fp8(torch.randn(m, k, device="cuda"))
+    bt = _fp8(torch.randn(n, k, device="cuda"))
+    ref = a.float() @ bt.float().t()
+    c = torch.empty(m, n, device="cuda", dtype=torch.float32)
+    attest.mfma_gemm_fp8(0, a.data_ptr(), bt.data_ptr(), c.data_ptr(), m, n, k)
+    torch.cuda.synchronize()
+    err = (c - ref).abs().max().item()
+    scale = ref.abs().max().item()
+    assert err <= 1e-3 * max(scale, 1.0), f"err={err} scale={scale}"
