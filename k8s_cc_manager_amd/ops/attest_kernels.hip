@@ -549,126 +549,144 @@ typedef __attribute__((ext_vector_type(4))) int v4i;
 constexpr int BK8 = 128;                 // fp8 elements per K-tile
 constexpr unsigned SCALE_ONE = 0x7F7F7F7Fu;  // e8m0 bias-127 = 2^0 per byte
 
-__device__ __forceinline__ v8i load_frag32(const char* p) {
-  v4i lo = *(const v4i*)(p);
-  v4i hi = *(const v4i*)(p + 16);
-  v8i out;
-  out[0] = lo[0]; out[1] = lo[1]; out[2] = lo[2]; out[3] = lo[3];
-  out[4] = hi[0]; out[5] = hi[1]; out[6] = hi[2]; out[7] = hi[3];
-  return out;
+// fp8 swizzle: XOR only byte bits 5-6 so every 32-B fragment stays
+// physically contiguous — the fragment can then be ONE v8i load into a
+// contiguous 8-register tuple (assembling it from two 16-B halves
+// leaves the allocator building 8-tuples out of scattered pairs, which
+// fragments past 256 VGPRs and spills). Costs a few 2-way LDS bank
+// conflicts; fp8's read count is half bf16's, so that is cheap.
+__device__ __forceinline__ int swz8(int byte_off) {
+  return byte_off ^ (((byte_off >> 8) & 3) << 5);
 }
 
-#define PHASE8(buf, mh, nh, ACC, PREFETCH_STMT, DRAIN, LOAD_A, LOAD_B)         \
-  do {                                                                         \
-    if (LOAD_A) {                                                              \
-      char* Ah = slot_ptr(lds, 0, (buf), (mh));                                \
-      _Pragma("unroll") for (int ks = 0; ks < 2; ++ks)                         \
-          _Pragma("unroll") for (int t = 0; t < 2; ++t) {                      \
-        int lg = (wave_mq + t * 32 + lane31) * 128 + ks * 64 + kq_b;           \
-        a8[t][ks] = load_frag32(Ah + swz(lg));                                 \
-      }                                                                        \
-    }                                                                          \
-    if (LOAD_B) {                                                              \
-      char* Bh = slot_ptr(lds, 1, (buf), (nh));                                \
-      _Pragma("unroll") for (int ks = 0; ks < 2; ++ks) {                       \
-        int lg = (wave_nq + lane31) * 128 + ks * 64 + kq_b;                    \
-        b8[ks] = load_frag32(Bh + swz(lg));                                    \
-      }                                                                        \
-    }                                                                          \
-    PREFETCH_STMT;                                                             \
-    DRAIN;                                                                     \
-    __builtin_amdgcn_s_barrier();                                              \
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                         \
-    __builtin_amdgcn_s_setprio(1);                                             \
-    _Pragma("unroll") for (int ks = 0; ks < 2; ++ks)                           \
-        _Pragma("unroll") for (int t = 0; t < 2; ++t)                          \
-            ACC[t] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(          \
-                a8[t][ks], b8[ks], ACC[t], 0, 0, 0, SCALE_ONE, 0, SCALE_ONE);  \
-    __builtin_amdgcn_s_setprio(0);                                             \
-    __builtin_amdgcn_s_barrier();                                              \
-  } while (0)
+__device__ __forceinline__ v8i load_frag32(const char* p) {
+  return *(const v8i*)(p);
+}
 
-__global__ __launch_bounds__(512, 1) void mfma_gemm_fp8_256(
+// staging for the fp8 image: same lane-linear glds, swz8 on the source
+__device__ __forceinline__ void stage_half_glds8(
+    const char* gbase, long row_stride_b, long k0_b, int row0,
+    char* lds_half, int wave, int lane) {
+#pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    int base = (p * 8 + wave) * 1024;
+    int logical = swz8(base + lane * 16);
+    int row = logical >> 7;
+    int colb = logical & 127;
+    const char* g = gbase + (long)(row0 + row) * row_stride_b + k0_b + colb;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)g,
+        (__attribute__((address_space(3))) void*)(lds_half + base), 16, 0, 0);
+  }
+}
+
+// ALL fragments are transient: loaded INSIDE the prio section with
+// one-MFMA lifetimes (only the 128-VGPR accumulators persist). At
+// 32 B/lane fragments, persisting A/B across the barrier — the bf16
+// template's shape — fragments the allocator around the 10-operand
+// scaled MFMA and spills glds addressing temporaries to scratch
+// (624-1052 B/lane measured), destroying the pipeline. Post-barrier
+// reads are safe: the compiler's counted lgkmcnt before each MFMA use
+// retires every read before the wave's last MFMA, hence before the
+// phase's closing barrier; the slot's refill DMA only issues after
+// that barrier.
+
+// 4-pass variant of the swz8 staging (16 KiB tile staged by 4 waves)
+__device__ __forceinline__ void stage_tile_glds8x4(
+    const char* gbase, long row_stride_b, long k0_b, char* lds_tile,
+    int wave, int lane) {
+#pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    int base = (wave * 4 + p) * 1024;
+    int logical = swz8(base + lane * 16);
+    int row = logical >> 7;
+    int colb = logical & 127;
+    const char* g = gbase + (long)row * row_stride_b + k0_b + colb;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)g,
+        (__attribute__((address_space(3))) void*)(lds_tile + base), 16, 0, 0);
+  }
+}
+
+// fp8 GEMM uses the PROVEN step-3 structure (the 128x128 tile of
+// mfma_gemm_bf16): 4 waves, each computing 64x64 as 4x4 16-tiles, one
+// 16x16x128 scaled MFMA per tile pair per K-chunk (BK=128 fp8 = one
+// instruction's worth), double-buffered glds staging, __syncthreads
+// between K-steps. The deep-pipelined 256-tile shape was tried and
+// REJECTED for fp8: 32-B/lane fragments push the allocator past 256
+// VGPRs, the spill's scratch traffic miscounts the hand-counted vmcnt
+// drains (scratch ops share the vm counter), and results corrupt. The
+// step-3 shape holds acc at 64 VGPRs, fragments at 64, and its
+// __syncthreads drain is immune to incidental VMEM ops. The CDNA4
+// guide's ladder measured MX-fp8 K=128 at 1628 TF/s in exactly this
+// structure.
+__global__ __launch_bounds__(256, 2) void mfma_gemm_fp8_128(
     const char* __restrict__ A, const char* __restrict__ Bt,
-    float* __restrict__ C, int M, int N, int K, int xcd_swizzle) {
-  __shared__ char lds[8 * HALF_B];
+    float* __restrict__ C, int M, int N, int K) {
+  __shared__ char lds[2 * 2 * 16384];  // [buf][A|B][16 KiB], one object
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
-  const int wave = tid >> 6;
-  const int wave_mq = (wave >> 2) * 64;
-  const int wave_nq = (wave & 3) * 32;
-  const int lane31 = lane & 31;
-  const int kq_b = (lane >> 5) * 32;  // fragment k byte offset (32 fp8)
-  int wg = blockIdx.y * gridDim.x + blockIdx.x;
-  if (xcd_swizzle) {
-    int nwg = gridDim.x * gridDim.y;
-    int q = nwg >> 3, r = nwg & 7;
-    int xcd = wg & 7, o = wg >> 3;
-    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + o;
-  }
-  const int block_m = (wg / gridDim.x) * BM2;
-  const int block_n = (wg % gridDim.x) * BN2;
+  const int wave = tid >> 6;            // 0..3
+  const int wave_m = (wave >> 1) * 64;
+  const int wave_n = (wave & 1) * 64;
+  const int block_m = blockIdx.y * BM;
+  const int block_n = blockIdx.x * BN;
 
   const char* gA = A + (long)block_m * K;
   const char* gB = Bt + (long)block_n * K;
-  const long row_b = (long)K;  // 1 byte per element
+  const long row_b = (long)K;  // bytes per row (1 B per element)
+
+  f32x4 acc[4][4] = {};  // 4x4 MFMA tiles of 16x16 per wave (64 VGPR)
+
+  const int lane15 = lane & 15;
+  const int kq_b = (lane >> 4) * 32;  // fragment k byte offset (32 fp8)
+
   const int nk = K / BK8;
-
-  f32x16 acc00[2] = {}, acc01[2] = {}, acc10[2] = {}, acc11[2] = {};
-  v8i a8[2][2], b8[2];
-
-#define STAGE(op, buf, half, tile)                                             \
-  stage_half_glds((op) == 0 ? gA : gB, row_b, (long)(tile) * BK8, (half) * 128,\
-                  slot_ptr(lds, (op), (buf), (half)), wave, lane)
-
-  STAGE(0, 0, 0, 0);
-  STAGE(1, 0, 0, 0);
-  STAGE(0, 0, 1, 0);
-  STAGE(1, 0, 1, 0);
-  STAGE(0, 1, 0, 1);
-  STAGE(1, 1, 1, 1);
-  STAGE(0, 1, 1, 1);
+  stage_tile_glds8x4(gA, row_b, 0, &lds[0], wave, lane);
+  stage_tile_glds8x4(gB, row_b, 0, &lds[16384], wave, lane);
   __syncthreads();
 
-#define VM_DRAIN                                                               \
-  do {                                                                         \
-    if (tp + 4 >= nk)                                                          \
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");                         \
-    else                                                                       \
-      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");                         \
-  } while (0)
-
-  for (int tp = 0; tp < nk; tp += 2) {
-    PHASE8(0, 0, 0, acc00, if (tp + 1 < nk) STAGE(1, 1, 0, tp + 1), , 1, 1);
-    PHASE8(0, 0, 1, acc01, if (tp + 2 < nk) STAGE(0, 0, 0, tp + 2), , 0, 1);
-    PHASE8(0, 1, 1, acc11, if (tp + 2 < nk) STAGE(1, 0, 1, tp + 2), , 1, 0);
-    PHASE8(0, 1, 0, acc10, if (tp + 2 < nk) STAGE(0, 0, 1, tp + 2), VM_DRAIN, 0, 1);
-    PHASE8(1, 0, 0, acc00, if (tp + 2 < nk) STAGE(1, 0, 0, tp + 2), , 1, 1);
-    PHASE8(1, 0, 1, acc01, if (tp + 3 < nk) STAGE(0, 1, 0, tp + 3), , 0, 1);
-    PHASE8(1, 1, 1, acc11, if (tp + 3 < nk) STAGE(1, 1, 1, tp + 3), , 1, 0);
-    PHASE8(1, 1, 0, acc10, if (tp + 3 < nk) STAGE(0, 1, 1, tp + 3), VM_DRAIN, 0, 1);
+  for (int kt = 0; kt < nk; ++kt) {
+    const int buf = kt & 1;
+    char* As = &lds[buf * 2 * 16384];
+    char* Bs = As + 16384;
+    if (kt + 1 < nk) {
+      char* An = &lds[(buf ^ 1) * 2 * 16384];
+      stage_tile_glds8x4(gA, row_b, (long)(kt + 1) * BK8, An, wave, lane);
+      stage_tile_glds8x4(gB, row_b, (long)(kt + 1) * BK8, An + 16384, wave, lane);
+    }
+    {
+      v8i afrag[4], bfrag[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int la = (wave_m + i * 16 + lane15) * 128 + kq_b;
+        int lb = (wave_n + i * 16 + lane15) * 128 + kq_b;
+        afrag[i] = load_frag32(As + swz8(la));
+        bfrag[i] = load_frag32(Bs + swz8(lb));
+      }
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+              afrag[i], bfrag[j], acc[i][j], 0, 0, 0, SCALE_ONE, 0, SCALE_ONE);
+    }
+    __syncthreads();
   }
-#undef VM_DRAIN
-#undef STAGE
 
-  const int c_col32 = lane & 31;
-  const int c_rowhi = (lane >> 5) * 4;
+  // epilogue: 16x16 C/D map col=lane&15, row=(lane>>4)*4+r
+  const int c_col = lane & 15;
+  const int c_row0 = (lane >> 4) * 4;
 #pragma unroll
-  for (int mh = 0; mh < 2; ++mh) {
+  for (int i = 0; i < 4; ++i) {
 #pragma unroll
-    for (int nh = 0; nh < 2; ++nh) {
-      f32x16* accq = mh == 0 ? (nh == 0 ? acc00 : acc01)
-                             : (nh == 0 ? acc10 : acc11);
+    for (int j = 0; j < 4; ++j) {
 #pragma unroll
-      for (int t = 0; t < 2; ++t) {
-#pragma unroll
-        for (int reg = 0; reg < 16; ++reg) {
-          int row = block_m + mh * 128 + wave_mq + t * 32 + (reg & 3) +
-                    8 * (reg >> 2) + c_rowhi;
-          int col = block_n + nh * 128 + wave_nq + c_col32;
-          C[(long)row * N + col] = accq[t][reg];
-        }
+      for (int r = 0; r < 4; ++r) {
+        int row = block_m + wave_m + i * 16 + c_row0 + r;
+        int col = block_n + wave_n + j * 16 + c_col;
+        C[(long)row * N + col] = acc[i][j][r];
       }
     }
   }
@@ -970,16 +988,14 @@ int cc_mfma_gemm_bf16_variant(int device, const void* A, const void* Bt,
 }
 
 // MX-scaled fp8 (e4m3) GEMM: C[M,N] = A[M,K] @ Bt[N,K]^T, fp8 inputs,
-// fp32 out. M,N multiples of 256; K multiple of 256.
+// fp32 out. M,N multiples of 128; K multiple of 128.
 int cc_mfma_gemm_fp8(int device, const void* A, const void* Bt, void* C,
                      int M, int N, int K) {
-  if (M % BM2 || N % BN2 || K % (2 * BK8)) return -2;
+  if (M % BM || N % BN || K % BK8) return -2;
   if (hipSetDevice(device) != hipSuccess) return -3;
-  dim3 grid(N / BN2, M / BM2);
-  long ws = (long)K * (M + N) + 4L * M * N;
-  hipLaunchKernelGGL(mfma_gemm_fp8_256, grid, dim3(512), 0, 0, (const char*)A,
-                     (const char*)Bt, (float*)C, M, N, K,
-                     ws > (256L << 20) ? 1 : 0);
+  dim3 grid(N / BN, M / BM);
+  hipLaunchKernelGGL(mfma_gemm_fp8_128, grid, dim3(256), 0, 0, (const char*)A,
+                     (const char*)Bt, (float*)C, M, N, K);
   return (int)hipDeviceSynchronize();
 }
 
