@@ -44,6 +44,9 @@ class _CReport(ctypes.Structure):
         ("ref_ms", ctypes.c_double),
         ("max_abs_err", ctypes.c_float),
         ("checksum", ctypes.c_ulonglong),
+        ("fp8_ms", ctypes.c_double),
+        ("fp8_tflops", ctypes.c_double),
+        ("fp8_max_abs_err", ctypes.c_float),
         ("lds_ms", ctypes.c_double),
         ("lds_failures", ctypes.c_uint),
         ("hbm_ms", ctypes.c_double),
@@ -66,6 +69,9 @@ class AttestReport:
     ref_ms: float
     max_abs_err: float
     checksum: int
+    fp8_ms: float
+    fp8_tflops: float
+    fp8_max_abs_err: float
     lds_ms: float
     lds_failures: int
     hbm_ms: float
@@ -88,6 +94,9 @@ class AttestReport:
             ref_ms=c.ref_ms,
             max_abs_err=c.max_abs_err,
             checksum=c.checksum,
+            fp8_ms=c.fp8_ms,
+            fp8_tflops=c.fp8_tflops,
+            fp8_max_abs_err=c.fp8_max_abs_err,
             lds_ms=c.lds_ms,
             lds_failures=c.lds_failures,
             hbm_ms=c.hbm_ms,
@@ -165,19 +174,20 @@ def attest_device(device_index: int, gemm_dim: int = 1024) -> AttestReport:
     if not rep.ok:
         raise AttestationError(
             f"device {device_index}: attestation FAILED "
-            f"(max_abs_err={rep.max_abs_err}, lds_failures={rep.lds_failures})"
+            f"(max_abs_err={rep.max_abs_err}, fp8_max_abs_err={rep.fp8_max_abs_err}, "
+            f"lds_failures={rep.lds_failures})"
         )
     logger.info(
-        "attested device %d: %s %d CUs, GEMM %.1f TF/s (%dx%dx%d, %.2f ms), "
-        "LDS ok, HBM %.0f GB/s, %d/%d xGMI peers",
+        "attested device %d: %s %d CUs, bf16 GEMM %.1f TF/s, fp8 GEMM "
+        "%.1f TF/s (%dx%dx%d), LDS ok, HBM %.0f GB/s, %d/%d xGMI peers",
         rep.device,
         rep.arch,
         rep.cu_count,
         rep.gemm_tflops,
+        rep.fp8_tflops,
         rep.gemm_dim,
         rep.gemm_dim,
         rep.gemm_dim,
-        rep.gemm_ms,
         rep.hbm_gbps,
         rep.peers_accessible,
         rep.peer_count,

@@ -74,6 +74,22 @@ __global__ void fill_bf16_lcg(bf16* __restrict__ out, long n, uint32_t seed) {
   }
 }
 
+// identical value sequence in OCP e4m3 (exact for {-2,-1,0,1}): lets
+// the fp8 MFMA path be verified BITWISE against the same fp32 VALU
+// reference the bf16 path uses.
+__global__ void fill_fp8_lcg(unsigned char* __restrict__ out, long n,
+                             uint32_t seed) {
+  const unsigned char enc[4] = {0xC0, 0xB8, 0x00, 0x38};  // -2,-1,0,1
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    uint32_t x = (uint32_t)i * 1103515245u + seed * 747796405u + 12345u;
+    x ^= x >> 16;
+    x *= 2654435769u;
+    out[i] = enc[(x >> 13) & 3u];
+  }
+}
+
 // ---------------------------------------------------------------------------
 // MFMA bf16 GEMM:  C[M,N] = A[M,K] @ Bt[N,K]^T      (all row-major)
 // Grid: (N/BN, M/BM); block: 256 threads (4 waves, 2x2 of 64x64).
@@ -805,6 +821,7 @@ constexpr int kMaxDevices = 64;
 struct ProbeCtx {
   int dim = 0;
   bf16 *dA = nullptr, *dB = nullptr;
+  unsigned char *dA8 = nullptr, *dB8 = nullptr;
   float *dC = nullptr, *dRef = nullptr, *dErr = nullptr;
   unsigned long long* dSum = nullptr;
   uint32_t* dFail = nullptr;
@@ -818,6 +835,8 @@ static std::mutex g_ctx_mu;
 static void ctx_release(ProbeCtx& c) {
   if (c.dA) (void)hipFree(c.dA);
   if (c.dB) (void)hipFree(c.dB);
+  if (c.dA8) (void)hipFree(c.dA8);
+  if (c.dB8) (void)hipFree(c.dB8);
   if (c.dC) (void)hipFree(c.dC);
   if (c.dRef) (void)hipFree(c.dRef);
   if (c.dErr) (void)hipFree(c.dErr);
@@ -840,6 +859,8 @@ static hipError_t ctx_acquire(int device, int dim, ProbeCtx** out) {
   hipError_t e;
   if ((e = hipMalloc(&c.dA, elems * sizeof(bf16))) != hipSuccess) return e;
   if ((e = hipMalloc(&c.dB, elems * sizeof(bf16))) != hipSuccess) return e;
+  if ((e = hipMalloc(&c.dA8, elems)) != hipSuccess) return e;
+  if ((e = hipMalloc(&c.dB8, elems)) != hipSuccess) return e;
   if ((e = hipMalloc(&c.dC, elems * sizeof(float))) != hipSuccess) return e;
   if ((e = hipMalloc(&c.dRef, elems * sizeof(float))) != hipSuccess) return e;
   if ((e = hipMalloc(&c.dErr, sizeof(float))) != hipSuccess) return e;
@@ -873,6 +894,10 @@ struct CcAttestReport {
   double ref_ms;
   float max_abs_err;       // MFMA vs VALU fp32 (must be 0.0: exact inputs)
   unsigned long long checksum;
+  // fp8 (MX-scaled e4m3) MFMA path, same integer data
+  double fp8_ms;
+  double fp8_tflops;
+  float fp8_max_abs_err;   // vs the same fp32 VALU reference (0.0)
   // LDS probe
   double lds_ms;
   unsigned int lds_failures;
@@ -1079,6 +1104,37 @@ int cc_attest_device(int device, int gemm_dim, struct CcAttestReport* rep) {
   CC_CHECK(hipMemcpy(&rep->checksum, dSum, sizeof(rep->checksum),
                      hipMemcpyDeviceToHost));
 
+  // -- fp8 (MX-scaled) MFMA path: same integer values, same fp32
+  // reference, bitwise requirement -------------------------------------
+  hipLaunchKernelGGL(fill_fp8_lcg, dim3(2048), dim3(256), 0, 0, ctx->dA8,
+                     elems, 1u);
+  hipLaunchKernelGGL(fill_fp8_lcg, dim3(2048), dim3(256), 0, 0, ctx->dB8,
+                     elems, 7u);
+  {
+    dim3 g8(D / BN, D / BM);
+    hipLaunchKernelGGL(mfma_gemm_fp8_128, g8, dim3(256), 0, 0,
+                       (const char*)ctx->dA8, (const char*)ctx->dB8, dC, D, D,
+                       D);
+    CC_CHECK(hipDeviceSynchronize());
+    CC_CHECK(hipEventRecord(ev0, 0));
+    hipLaunchKernelGGL(mfma_gemm_fp8_128, g8, dim3(256), 0, 0,
+                       (const char*)ctx->dA8, (const char*)ctx->dB8, dC, D, D,
+                       D);
+    CC_CHECK(hipEventRecord(ev1, 0));
+    CC_CHECK(hipEventSynchronize(ev1));
+    rep->fp8_ms = event_ms(ev0, ev1);
+    rep->fp8_tflops = 2.0 * D * (double)D * D / (rep->fp8_ms * 1e-3) / 1e12;
+    CC_CHECK(hipMemset(dErr, 0, sizeof(float)));
+    hipLaunchKernelGGL(max_abs_diff, dim3(1024), dim3(256), 0, 0, dC, dRef,
+                       elems, dErr);
+    CC_CHECK(hipDeviceSynchronize());
+    unsigned int e8 = 0;
+    CC_CHECK(hipMemcpy(&e8, dErr, sizeof(e8), hipMemcpyDeviceToHost));
+    float f8err;
+    __builtin_memcpy(&f8err, &e8, sizeof(f8err));
+    rep->fp8_max_abs_err = f8err;
+  }
+
   // -- LDS probe -------------------------------------------------------
   uint32_t* dFail = ctx->dFail;
   CC_CHECK(hipMemset(dFail, 0, sizeof(uint32_t)));
@@ -1111,8 +1167,8 @@ int cc_attest_device(int device, int gemm_dim, struct CcAttestReport* rep) {
       ++rep->peers_accessible;
   }
 
-  rep->ok = (rep->max_abs_err == 0.0f) && (rep->lds_failures == 0) &&
-            (rep->gemm_tflops > 0.0) ? 1 : 0;
+  rep->ok = (rep->max_abs_err == 0.0f) && (rep->fp8_max_abs_err == 0.0f) &&
+            (rep->lds_failures == 0) && (rep->gemm_tflops > 0.0) ? 1 : 0;
   return 0;
 }
 

@@ -112,8 +112,10 @@ def test_attest_device_full_probe(attest):
     assert rep.ok
     assert "gfx950" in rep.arch or rep.cu_count > 0
     assert rep.max_abs_err == 0.0  # integer inputs: bitwise agreement
+    assert rep.fp8_max_abs_err == 0.0  # fp8 MFMA path, same ground truth
     assert rep.lds_failures == 0
     assert rep.gemm_tflops > 10.0, f"MFMA path suspiciously slow: {rep.gemm_tflops}"
+    assert rep.fp8_tflops > 10.0, f"fp8 MFMA path suspiciously slow: {rep.fp8_tflops}"
     assert rep.hbm_gbps > 500.0
     assert rep.checksum != 0
 
