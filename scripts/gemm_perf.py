@@ -22,8 +22,27 @@ def bench_size(n: int, iters: int = 20) -> dict:
     bt = torch.randn(n, n, device="cuda", dtype=torch.float32).bfloat16()
     c = torch.empty(n, n, device="cuda", dtype=torch.float32)
     out = {"n": n}
-    for which, name in ((0, "v128"), (1, "v256"), (2, "v256w")):
-        if which >= 1 and (n % 256 or n % 128):
+    for which, name in ((0, "v128"), (1, "v256"), (2, "v256w"), (3, "fp8")):
+        if which in (1, 2) and (n % 256 or n % 128):
+            continue
+        if which == 3:
+            if n % 128:
+                continue
+            a8 = torch.randn(n, n, device="cuda").to(torch.float8_e4m3fn)
+            bt8 = torch.randn(n, n, device="cuda").to(torch.float8_e4m3fn)
+            run = lambda: attest.mfma_gemm_fp8(  # noqa: E731
+                0, a8.data_ptr(), bt8.data_ptr(), c.data_ptr(), n, n, n
+            )
+            run()
+            torch.cuda.synchronize()
+            t0 = time.perf_counter()
+            for _ in range(iters):
+                run()
+            torch.cuda.synchronize()
+            dt = (time.perf_counter() - t0) / iters
+            out["fp8_ms"] = round(dt * 1e3, 3)
+            out["fp8_tflops"] = round(2.0 * n * n * n / dt / 1e12, 1)
+            del a8, bt8
             continue
         run = lambda: attest.mfma_gemm_bf16_variant(  # noqa: E731
             0, a.data_ptr(), bt.data_ptr(), c.data_ptr(), n, n, n, which

@@ -1,0 +1,74 @@
+"""DeviceExecutor unit tests: aggregation, partition order, single-worker path."""
+
+import threading
+
+import pytest
+
+from k8s_cc_manager_amd.device.mock import MockBackend
+from k8s_cc_manager_amd.parallel.executor import DeviceExecutor, PerDeviceError
+
+
+def test_run_collects_all_results():
+    be = MockBackend(num_gpus=4)
+    devices, _ = be.find_devices()
+    ex = DeviceExecutor()
+    out = ex.run("query", devices, lambda d: d.query_cc_mode())
+    assert set(out) == {d.bdf for d in devices}
+    assert all(v == "off" for v in out.values())
+
+
+def test_run_aggregates_every_failure():
+    be = MockBackend(num_gpus=4)
+    devices, _ = be.find_devices()
+    bad = {devices[1].bdf, devices[3].bdf}
+
+    def fn(d):
+        if d.bdf in bad:
+            raise RuntimeError(f"boom {d.bdf}")
+        return "ok"
+
+    with pytest.raises(PerDeviceError) as ei:
+        DeviceExecutor().run("phase-x", devices, fn)
+    err = ei.value
+    assert err.phase == "phase-x"
+    assert set(err.errors) == bad
+    # message names every failing device
+    for bdf in bad:
+        assert bdf in str(err)
+
+
+def test_partition_preserves_device_order():
+    be = MockBackend(num_gpus=6)
+    devices, _ = be.find_devices()
+    yes, no = DeviceExecutor().partition(
+        "p", devices, lambda d: devices.index(d) % 2 == 0
+    )
+    assert [devices.index(d) for d in yes] == [0, 2, 4]
+    assert [devices.index(d) for d in no] == [1, 3, 5]
+
+
+def test_empty_device_list():
+    assert DeviceExecutor().run("p", [], lambda d: 1) == {}
+
+
+def test_single_device_runs_inline():
+    """One device must not spawn a pool (runs on the calling thread)."""
+    be = MockBackend(num_gpus=1)
+    devices, _ = be.find_devices()
+    caller = threading.current_thread().name
+    seen = []
+    DeviceExecutor().run("p", devices, lambda d: seen.append(threading.current_thread().name))
+    assert seen == [caller]
+
+
+def test_concurrency_actually_parallel():
+    be = MockBackend(num_gpus=8)
+    devices, _ = be.find_devices()
+    barrier = threading.Barrier(8, timeout=10)
+
+    def fn(d):
+        barrier.wait()  # deadlocks unless all 8 run concurrently
+        return True
+
+    out = DeviceExecutor().run("p", devices, fn)
+    assert len(out) == 8

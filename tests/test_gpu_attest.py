@@ -61,7 +61,9 @@ def test_mfma_gemm_integer_exact(attest):
 
 
 @pytest.mark.parametrize("which", [1, 2])
-@pytest.mark.parametrize("m,n,k", [(256, 256, 128), (512, 256, 384), (256, 512, 1024)])
+@pytest.mark.parametrize(
+    "m,n,k", [(256, 256, 128), (512, 256, 384), (256, 256, 512), (256, 512, 1024)]
+)
 def test_mfma_gemm_256_template_matches_torch(attest, m, n, k, which):
     """The deep-pipelined 256x256 8-phase variant — race-sensitive, so
     run it several times per shape (sync-structure screening)."""
