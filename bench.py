@@ -44,6 +44,14 @@ def parse_args(argv=None):
     p.add_argument("--warmup", type=int, default=2, help="untimed warmup steps")
     p.add_argument("--attest-dim", type=int, default=1024, help="attestation GEMM size")
     p.add_argument("--mock", action="store_true", help="force the mock device tier")
+    p.add_argument(
+        "--device-backend",
+        choices=["shadow", "amdsmi", "mock"],
+        default="",
+        help="device tier override: shadow (default on GPU; safe), amdsmi "
+        "(real backend — FLR only with CC_MANAGER_ALLOW_RESET=1, i.e. "
+        "BASELINE config 2 on dedicated hardware), mock",
+    )
     p.add_argument("--no-evict", action="store_true", help="skip the eviction leg")
     p.add_argument(
         "--workload",
@@ -93,8 +101,29 @@ def main(argv=None) -> int:
     k8s = K8sClient(url)
 
     # ---- device tier --------------------------------------------------
+    if args.mock:
+        tier = "mock"
+    elif args.device_backend:
+        tier = args.device_backend
+    else:
+        tier = "shadow" if torch.cuda.is_available() else "mock"
+    use_gpu = use_gpu and tier != "mock"
     n_managed = args.gpus if world == 1 else 1
-    if use_gpu:
+    if tier == "amdsmi" and use_gpu:
+        from k8s_cc_manager_amd.device.amdsmi_backend import AmdSmiBackend
+        from k8s_cc_manager_amd.ops import attest
+
+        backend_dev = AmdSmiBackend()
+        if world > 1:  # one rank manages one GPU
+            backend_dev._devices = [backend_dev._devices[local_rank]]
+        indices = [local_rank]
+        attestor = lambda dev: attest.attest_device(  # noqa: E731
+            max(dev.hip_index, 0), gemm_dim=args.attest_dim
+        )
+        device_tier = "amdsmi" + (
+            "+flr" if os.environ.get("CC_MANAGER_ALLOW_RESET") == "1" else "-shadowreg"
+        )
+    elif use_gpu:
         from k8s_cc_manager_amd.device.shadow import ShadowBackend
         from k8s_cc_manager_amd.ops import attest
 
