@@ -20,7 +20,6 @@ from .contract import (
     CCDevice,
     DeviceBackend,
     FABRIC_OFF,
-    ModeVerifyError,
     ResetError,
 )
 
