@@ -624,6 +624,142 @@ __device__ __forceinline__ void stage_tile_glds8x4(
   }
 }
 
+// Deep-pipelined fp8: the same 8-phase 256x256 schedule as the bf16
+// template (identical 16 KiB half-tiles, prefetch map and vmcnt(6)
+// drains — the fp8 image has the same 128-B rows: 128 fp8 elements per
+// K-tile), with the scaled MFMA issued via INLINE ASM. The
+// __builtin_amdgcn_mfma_scale_* intrinsic's register-class constraints
+// interact pathologically with global_load_lds on ROCm 7.2: the
+// identical structure compiles to 180 VGPR/no-spill with bf16 MFMA but
+// 256 VGPR + ~1 KB/lane scratch with the intrinsic (and scratch ops
+// share the vm counter, corrupting counted-vmcnt drains). The asm form
+// compiles clean (162 VGPR in the bisect kernel). Hazards: inputs are
+// ds_read results (compiler inserts lgkm waits for asm data deps);
+// MFMA->MFMA on one accumulator is pipe-ordered; the only uncovered
+// RAW is accumulator->epilogue VALU, guarded by the post-loop s_nops.
+#define MFMA_FP8_ASM(ACCX, AOP, BOP)                                          \
+  asm volatile(                                                               \
+      "v_mfma_scale_f32_16x16x128_f8f6f4 %0, %1, %2, %0, %3, %4 "             \
+      "op_sel_hi:[0,0,0]"                                                     \
+      : "+v"(ACCX)                                                            \
+      : "v"(AOP), "v"(BOP), "v"(sc_reg), "v"(sc_reg))
+
+#define PHASE8D(buf, mh, nh, LOAD_A, LOAD_B, ACC, PREFETCH_STMT, DRAIN)       \
+  do {                                                                        \
+    if (LOAD_A) {                                                             \
+      char* Ah = slot_ptr(lds, 0, (buf), (mh));                               \
+      _Pragma("unroll") for (int i = 0; i < 4; ++i) {                         \
+        int lg = (wave_mq + i * 16 + lane15) * 128 + kq_b;                    \
+        a8[i] = *(const v8i*)(Ah + swz8(lg));                                 \
+      }                                                                       \
+    }                                                                         \
+    if (LOAD_B) {                                                             \
+      char* Bh = slot_ptr(lds, 1, (buf), (nh));                               \
+      _Pragma("unroll") for (int j = 0; j < 2; ++j) {                         \
+        int lg = (wave_nq + j * 16 + lane15) * 128 + kq_b;                    \
+        b8[j] = *(const v8i*)(Bh + swz8(lg));                                 \
+      }                                                                       \
+    }                                                                         \
+    PREFETCH_STMT;                                                            \
+    DRAIN;                                                                    \
+    __builtin_amdgcn_s_barrier();                                             \
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                        \
+    __builtin_amdgcn_s_setprio(1);                                            \
+    _Pragma("unroll") for (int i = 0; i < 4; ++i)                             \
+        _Pragma("unroll") for (int j = 0; j < 2; ++j)                         \
+            MFMA_FP8_ASM(ACC[i][j], a8[i], b8[j]);                            \
+    __builtin_amdgcn_s_setprio(0);                                            \
+    __builtin_amdgcn_s_barrier();                                             \
+  } while (0)
+
+__global__ __launch_bounds__(512, 1) void mfma_gemm_fp8_256(
+    const char* __restrict__ A, const char* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K, int xcd_swizzle) {
+  __shared__ char lds[8 * HALF_B];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wave_mq = (wave >> 2) * 64;
+  const int wave_nq = (wave & 3) * 32;
+  const int lane15 = lane & 15;
+  const int kq_b = (lane >> 4) * 32;
+  int wg = blockIdx.y * gridDim.x + blockIdx.x;
+  if (xcd_swizzle) {
+    int nwg = gridDim.x * gridDim.y;
+    int q = nwg >> 3, r = nwg & 7;
+    int xcd = wg & 7, o = wg >> 3;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + o;
+  }
+  const int block_m = (wg / gridDim.x) * BM2;
+  const int block_n = (wg % gridDim.x) * BN2;
+
+  const char* gA = A + (long)block_m * K;
+  const char* gB = Bt + (long)block_n * K;
+  const long row_b = (long)K;
+  const int nk = K / BK8;
+
+  f32x4 acc00[4][2] = {}, acc01[4][2] = {}, acc10[4][2] = {}, acc11[4][2] = {};
+  v8i a8[4], b8[2];
+  int sc_reg;
+  asm("v_mov_b32 %0, 0x7f7f7f7f" : "=v"(sc_reg));  // e8m0 1.0 x4
+
+#define STAGE(op, buf, half, tile)                                            \
+  stage_half_glds8((op) == 0 ? gA : gB, row_b, (long)(tile) * BK8,            \
+                   (half) * 128, slot_ptr(lds, (op), (buf), (half)), wave,    \
+                   lane)
+
+  STAGE(0, 0, 0, 0);
+  STAGE(1, 0, 0, 0);
+  STAGE(0, 0, 1, 0);
+  STAGE(1, 0, 1, 0);
+  STAGE(0, 1, 0, 1);
+  STAGE(1, 1, 1, 1);
+  STAGE(0, 1, 1, 1);
+  __syncthreads();
+
+#define VM_DRAIN                                                              \
+  do {                                                                        \
+    if (tp + 4 >= nk)                                                         \
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");                        \
+    else                                                                      \
+      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");                        \
+  } while (0)
+
+  for (int tp = 0; tp < nk; tp += 2) {
+    PHASE8D(0, 0, 0, 1, 1, acc00, if (tp + 1 < nk) STAGE(1, 1, 0, tp + 1), );
+    PHASE8D(0, 0, 1, 0, 1, acc01, if (tp + 2 < nk) STAGE(0, 0, 0, tp + 2), );
+    PHASE8D(0, 1, 1, 1, 0, acc11, if (tp + 2 < nk) STAGE(1, 0, 1, tp + 2), );
+    PHASE8D(0, 1, 0, 0, 1, acc10, if (tp + 2 < nk) STAGE(0, 0, 1, tp + 2), VM_DRAIN);
+    PHASE8D(1, 0, 0, 1, 1, acc00, if (tp + 2 < nk) STAGE(1, 0, 0, tp + 2), );
+    PHASE8D(1, 0, 1, 0, 1, acc01, if (tp + 3 < nk) STAGE(0, 1, 0, tp + 3), );
+    PHASE8D(1, 1, 1, 1, 0, acc11, if (tp + 3 < nk) STAGE(1, 1, 1, tp + 3), );
+    PHASE8D(1, 1, 0, 0, 1, acc10, if (tp + 3 < nk) STAGE(0, 1, 1, tp + 3), VM_DRAIN);
+  }
+#undef VM_DRAIN
+#undef STAGE
+
+  // accumulator -> VALU RAW guard (the compiler cannot see through the
+  // asm MFMAs; ~34 cycles of nops once per kernel)
+  asm volatile("s_nop 15\ns_nop 15\ns_nop 2" :::);
+
+  const int c_col = lane & 15;
+  const int c_row0 = (lane >> 4) * 4;
+#define EPI(ACC, mh, nh)                                                      \
+  _Pragma("unroll") for (int i = 0; i < 4; ++i)                               \
+      _Pragma("unroll") for (int j = 0; j < 2; ++j)                           \
+      _Pragma("unroll") for (int r = 0; r < 4; ++r) {                         \
+    int row = block_m + (mh)*128 + wave_mq + i * 16 + c_row0 + r;             \
+    int col = block_n + (nh)*128 + wave_nq + j * 16 + c_col;                  \
+    C[(long)row * N + col] = ACC[i][j][r];                                    \
+  }
+  EPI(acc00, 0, 0)
+  EPI(acc01, 0, 1)
+  EPI(acc10, 1, 0)
+  EPI(acc11, 1, 1)
+#undef EPI
+}
+
 // fp8 GEMM uses the PROVEN step-3 structure (the 128x128 tile of
 // mfma_gemm_bf16): 4 waves, each computing 64x64 as 4x4 16-tiles, one
 // 16x16x128 scaled MFMA per tile pair per K-chunk (BK=128 fp8 = one
@@ -1016,8 +1152,16 @@ int cc_mfma_gemm_bf16_variant(int device, const void* A, const void* Bt,
 // fp32 out. M,N multiples of 128; K multiple of 128.
 int cc_mfma_gemm_fp8(int device, const void* A, const void* Bt, void* C,
                      int M, int N, int K) {
-  if (M % BM || N % BN || K % BK8) return -2;
   if (hipSetDevice(device) != hipSuccess) return -3;
+  if (M % BM2 == 0 && N % BN2 == 0 && K % (2 * BK8) == 0) {
+    dim3 grid(N / BN2, M / BM2);
+    long ws = (long)K * (M + N) + 4L * M * N;
+    hipLaunchKernelGGL(mfma_gemm_fp8_256, grid, dim3(512), 0, 0,
+                       (const char*)A, (const char*)Bt, (float*)C, M, N, K,
+                       ws > (256L << 20) ? 1 : 0);
+    return (int)hipDeviceSynchronize();
+  }
+  if (M % BM || N % BN || K % BK8) return -2;
   dim3 grid(N / BN, M / BM);
   hipLaunchKernelGGL(mfma_gemm_fp8_128, grid, dim3(256), 0, 0, (const char*)A,
                      (const char*)Bt, (float*)C, M, N, K);

@@ -38,6 +38,33 @@ def test_fp8_gemm_integer_exact(attest):
         assert torch.equal(c, ref), f"trial {trial}: max diff {(c-ref).abs().max()}"
 
 
+def test_fp8_deep_kernel_race_screen(attest):
+    """The deep-pipelined fp8 kernel (M,N%256, K%256): bitwise integer
+    race screen across repeated runs (sync-structure discipline)."""
+    m = n = 512
+    k = 768  # K%256 == 0 -> deep kernel; odd nk exercises tail drains
+    a = _fp8(torch.randint(-2, 2, (m, k), device="cuda").float())
+    bt = _fp8(torch.randint(-2, 2, (n, k), device="cuda").float())
+    ref = a.float() @ bt.float().t()
+    for trial in range(10):
+        c = torch.empty(m, n, device="cuda", dtype=torch.float32)
+        attest.mfma_gemm_fp8(0, a.data_ptr(), bt.data_ptr(), c.data_ptr(), m, n, k)
+        torch.cuda.synchronize()
+        assert torch.equal(c, ref), f"trial {trial}"
+
+
+def test_fp8_step3_kernel_still_used_for_small_shapes(attest):
+    """K%256 != 0 routes to the 128-tile step-3 kernel."""
+    m, n, k = 128, 128, 384
+    a = _fp8(torch.randint(-2, 2, (m, k), device="cuda").float())
+    bt = _fp8(torch.randint(-2, 2, (n, k), device="cuda").float())
+    ref = a.float() @ bt.float().t()
+    c = torch.empty(m, n, device="cuda", dtype=torch.float32)
+    attest.mfma_gemm_fp8(0, a.data_ptr(), bt.data_ptr(), c.data_ptr(), m, n, k)
+    torch.cuda.synchronize()
+    assert torch.equal(c, ref)
+
+
 @pytest.mark.parametrize("m,n,k", [(256, 512, 256), (512, 256, 1024)])
 def test_fp8_gemm_random_matches_reference(attest, m, n, k):
     torch.manual_seed(m + n + k)
