@@ -1,0 +1,111 @@
+"""Node diagnostics: ``python -m k8s_cc_manager_amd.doctor``.
+
+One JSON document answering the first questions an operator asks on a
+misbehaving CC node: can the host do TEE at all, is KFD alive, what do
+PCI and amdsmi each see, does the attestation library load, and (with
+``--attest``) does a full probe pass. Every section degrades gracefully
+— a CPU-only box reports absence, not a stack trace.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import sys
+from typing import Any, Dict
+
+from .core.hostprobe import is_host_cc_enabled
+
+
+def collect(run_attest: bool = False, gemm_dim: int = 512) -> Dict[str, Any]:
+    report: Dict[str, Any] = {"schema": "cc-doctor/v1"}
+
+    # host TEE capability
+    report["host_cc_enabled"] = is_host_cc_enabled()
+
+    # native library + KFD + PCI
+    native_info: Dict[str, Any] = {"available": False}
+    try:
+        from .device import native
+
+        native_info["available"] = native.available()
+        if native_info["available"]:
+            ver = native.kfd_version()
+            native_info["kfd_version"] = list(ver) if ver else None
+            topo = [e for e in native.kfd_topology() if e.get("is_gpu")]
+            native_info["kfd_gpu_nodes"] = [
+                {
+                    "node": e.get("node"),
+                    "gfx_target_version": e.get("gfx_target_version"),
+                    "cu_count": e.get("cu_count"),
+                    "io_links_count": e.get("io_links_count"),
+                }
+                for e in topo
+            ]
+            native_info["pci_amd_gpus"] = native.pci_scan()
+    except Exception as e:
+        native_info["error"] = str(e)
+    report["native"] = native_info
+
+    # amdsmi enumeration
+    smi: Dict[str, Any] = {"available": False}
+    try:
+        from .device.amdsmi_backend import AmdSmiBackend
+
+        be = AmdSmiBackend()
+        devices, count = be.find_devices()
+        smi["available"] = True
+        smi["devices"] = [
+            {"bdf": d.bdf, "name": d.name, "cc_mode": d.query_cc_mode(),
+             "fabric_mode": d.query_fabric_mode()}
+            for d in devices
+        ]
+        smi["count"] = count
+    except Exception as e:
+        smi["error"] = str(e)
+    report["amdsmi"] = smi
+
+    # attestation library
+    att: Dict[str, Any] = {"library_loaded": False}
+    try:
+        from .ops import attest
+
+        attest._load()
+        att["library_loaded"] = True
+        att["hip_device_count"] = attest.device_count()
+        if run_attest and att["hip_device_count"] > 0:
+            import dataclasses
+
+            reports = []
+            for i in range(att["hip_device_count"]):
+                rep = attest.attest_device(i, gemm_dim=gemm_dim)
+                reports.append(dataclasses.asdict(rep))
+            att["probes"] = reports
+    except Exception as e:
+        att["error"] = str(e)
+    report["attestation"] = att
+
+    # verdict
+    ok_gpu = att.get("hip_device_count", 0) > 0 and att.get("library_loaded")
+    report["verdict"] = {
+        "cc_capable": bool(report["host_cc_enabled"] and ok_gpu),
+        "notes": []
+        + ([] if report["host_cc_enabled"] else ["host TEE (SEV-SNP/TDX) disabled"])
+        + ([] if ok_gpu else ["no attestable GPU (library or device missing)"]),
+    }
+    return report
+
+
+def main(argv=None) -> int:
+    ap = argparse.ArgumentParser(prog="cc-doctor", description=__doc__)
+    ap.add_argument("--attest", action="store_true",
+                    help="run the full attestation probe on every GPU")
+    ap.add_argument("--gemm-dim", type=int, default=512)
+    args = ap.parse_args(argv)
+    report = collect(run_attest=args.attest, gemm_dim=args.gemm_dim)
+    print(json.dumps(report, indent=2))
+    return 0 if report["verdict"]["cc_capable"] or not args.attest else 1
+
+
+if __name__ == "__main__":
+    sys.exit(main())
