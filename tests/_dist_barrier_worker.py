@@ -23,6 +23,11 @@ def main() -> None:
     report = engine.apply_cc_mode(devices, devices, "on")
     assert report.ok, report.error
     assert be.device(0).query_cc_mode() == "on"
+    # fabric-wide (ppcie) transition across ranks: the xGMI-hive
+    # stage-all/reset-all seam synchronizes via the same barrier
+    report = engine.apply_fabric_mode(devices)
+    assert report.ok, report.error
+    assert be.device(0).query_fabric_mode() == "on"
     dist.barrier()
     dist.destroy_process_group()
     print("RANK_OK", os.environ.get("RANK"))

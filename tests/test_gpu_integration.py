@@ -73,6 +73,26 @@ def test_watch_loop_with_real_attestation(fake_cluster, tmp_path, monkeypatch):
     assert rec["ok"] and rec["max_abs_err"] == 0.0 and rec["fp8_max_abs_err"] == 0.0
 
 
+def test_transition_with_deep_attestation(tmp_path, monkeypatch):
+    """CC_ATTEST_DEEP=1: the transition's verify phase re-runs the probe
+    under rocprofv3 and gates on hardware MFMA counters."""
+    from k8s_cc_manager_amd.core.transition import TransitionEngine
+    from k8s_cc_manager_amd.device.shadow import ShadowBackend
+    from k8s_cc_manager_amd.ops.attest import attest_device_by_bdf
+    from k8s_cc_manager_amd.ops.deep_attest import rocprof_available
+
+    if not rocprof_available():
+        pytest.skip("rocprofv3 not installed")
+    monkeypatch.setenv("CC_ATTEST_DEEP", "1")
+    monkeypatch.setenv("CC_ATTEST_GEMM_DIM", "512")
+    be = ShadowBackend(device_indices=[0])
+    devices, _ = be.find_devices()
+    engine = TransitionEngine(attestor=attest_device_by_bdf, boot_timeout=30)
+    report = engine.apply_cc_mode(devices, devices, "on")
+    assert report.ok, report.error
+    assert report.phases["verify"] > 1.0  # rocprof session is seconds
+
+
 def test_bench_amdsmi_tier(tmp_path):
     """bench --device-backend amdsmi (no FLR without CC_MANAGER_ALLOW_RESET)."""
     out = tmp_path / "b.json"
