@@ -154,13 +154,21 @@ def main(argv=None) -> int:
         engine=engine,
         config=ManagerConfig(
             evict_components=not args.no_evict,
+            evict_gpu_workloads=args.workload,
             cordon_node=True,
             eviction_timeout=30.0,
             eviction_poll_interval=0.01,
         ),
     )
+    if args.workload:
+        # a synthetic GPU workload pod: evicted (Eviction API) during
+        # each transition and readmitted by the "controller" (us) after
+        cluster.add_pod("user-ns", "synthetic-train", node_name,
+                        app="synthetic", gpu_request=1)
 
     # ---- optional synthetic workload (config 5: toggle under load) ----
+    # the workload POD (evict + readmit) is active on every tier; the
+    # GPU-contention thread additionally runs when a GPU is present
     workload_stop = None
     workload_thread = None
     if args.workload and use_gpu:
@@ -190,6 +198,9 @@ def main(argv=None) -> int:
         ok = manager.apply_mode(manager.with_default(label))
         if not ok:
             raise RuntimeError(f"rank {rank}: reconcile step {i} failed")
+        if args.workload:  # controller readmits the evicted workload pod
+            cluster.add_pod("user-ns", "synthetic-train", node_name,
+                            app="synthetic", gpu_request=1)
 
     def sync() -> None:
         if use_gpu:
@@ -245,7 +256,7 @@ def main(argv=None) -> int:
                 "eviction": not args.no_evict,
                 "components": len(COMPONENT_LABELS),
                 "attest_gemm_dim": args.attest_dim if use_gpu else 0,
-                "workload": bool(args.workload and use_gpu),
+                "workload": bool(args.workload),
                 "sec_per_gpu_transition": round(elapsed / args.steps, 4),
             },
         }

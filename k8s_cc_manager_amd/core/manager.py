@@ -47,6 +47,9 @@ class FatalConfigError(Exception):
 class ManagerConfig:
     operator_namespace: str = "amd-gpu-operator"
     evict_components: bool = True
+    #: also evict user pods requesting amd.com/gpu via the Eviction API
+    #: (beyond parity: the reference only pauses operator components)
+    evict_gpu_workloads: bool = False
     cordon_node: bool = True
     eviction_timeout: float = 300.0
     eviction_poll_interval: float = 2.0
@@ -63,6 +66,8 @@ class ManagerConfig:
         return cls(
             operator_namespace=env.get("OPERATOR_NAMESPACE", "amd-gpu-operator"),
             evict_components=env.get("EVICT_OPERATOR_COMPONENTS", "true").lower()
+            == "true",
+            evict_gpu_workloads=env.get("EVICT_GPU_WORKLOADS", "false").lower()
             == "true",
             cordon_node=env.get("CORDON_NODE", "true").lower() == "true",
             readiness_file=env.get("CC_READINESS_FILE"),
@@ -256,6 +261,14 @@ class CCManager:
             if cordoned:
                 eviction.uncordon(self.k8s, self.node_name)
             return False
+
+        if cfg.evict_gpu_workloads:
+            eviction.evict_gpu_workload_pods(
+                self.k8s,
+                self.node_name,
+                timeout=cfg.eviction_timeout,
+                poll_interval=cfg.eviction_poll_interval,
+            )
 
         ok = self._run_direct(mode, runner)
 

@@ -165,11 +165,29 @@ class K8sClient:
             params["fieldSelector"] = field_selector
         if label_selector:
             params["labelSelector"] = label_selector
-        resp = self._session.get(
-            f"{self.base_url}/api/v1/namespaces/{namespace}/pods", params=params
+        url = (
+            f"{self.base_url}/api/v1/namespaces/{namespace}/pods"
+            if namespace
+            else f"{self.base_url}/api/v1/pods"  # all namespaces
         )
+        resp = self._session.get(url, params=params)
         _raise_for(resp)
         return resp.json()
+
+    def evict_pod(self, namespace: str, name: str) -> None:
+        """Graceful eviction via the pods/eviction subresource (respects
+        PodDisruptionBudgets, unlike a raw DELETE)."""
+        body = {
+            "apiVersion": "policy/v1",
+            "kind": "Eviction",
+            "metadata": {"name": name, "namespace": namespace},
+        }
+        resp = self._session.post(
+            f"{self.base_url}/api/v1/namespaces/{namespace}/pods/{name}/eviction",
+            data=json.dumps(body),
+            headers={"Content-Type": "application/json"},
+        )
+        _raise_for(resp)
 
     # -- watch ----------------------------------------------------------
     def watch_node(

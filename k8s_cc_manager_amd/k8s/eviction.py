@@ -188,6 +188,79 @@ def unwind_paused_labels(
         logger.error("could not unwind paused labels: %s", e)
 
 
+GPU_RESOURCE = "amd.com/gpu"
+
+
+def _requests_gpu(pod: dict) -> bool:
+    for ctr in (pod.get("spec") or {}).get("containers") or []:
+        res = (ctr.get("resources") or {})
+        for kind in ("requests", "limits"):
+            if GPU_RESOURCE in (res.get(kind) or {}):
+                return True
+    return False
+
+
+def evict_gpu_workload_pods(
+    k8s: K8sClient,
+    node_name: str,
+    timeout: float = 300.0,
+    poll_interval: float = 2.0,
+    skip_namespaces: tuple = ("kube-system",),
+) -> bool:
+    """Evict every pod on the node that requests ``amd.com/gpu``, via
+    the pods/eviction subresource, and wait for them to terminate.
+
+    Beyond-parity capability (the reference only pauses operator
+    components): an FLR kills any process holding the device, so user
+    GPU workloads must be off the node before the reset — this is the
+    'evict + readmit under load' path of BASELINE config 5. Opt-in via
+    ``EVICT_GPU_WORKLOADS=true``; controllers (Deployments/Jobs)
+    recreate the pods elsewhere or after uncordon.
+    """
+    try:
+        pods = k8s.list_pods("", field_selector=f"spec.nodeName={node_name}")
+    except ApiError as e:
+        logger.error("could not list pods for GPU-workload eviction: %s", e)
+        return False
+    targets = [
+        (p["metadata"]["namespace"], p["metadata"]["name"])
+        for p in pods.get("items") or []
+        if _requests_gpu(p) and p["metadata"]["namespace"] not in skip_namespaces
+    ]
+    if not targets:
+        return True
+    logger.info("evicting %d GPU workload pod(s): %s", len(targets),
+                [f"{ns}/{n}" for ns, n in targets])
+    ok = True
+    for ns, name in targets:
+        try:
+            k8s.evict_pod(ns, name)
+        except ApiError as e:
+            logger.warning("eviction of %s/%s rejected: %s", ns, name, e)
+            ok = False
+
+    deadline = time.monotonic() + timeout
+    delay = min(0.002, poll_interval)
+    remaining = set(targets)
+    while remaining and time.monotonic() < deadline:
+        try:
+            pods = k8s.list_pods("", field_selector=f"spec.nodeName={node_name}")
+            alive = {
+                (p["metadata"]["namespace"], p["metadata"]["name"])
+                for p in pods.get("items") or []
+            }
+            remaining &= alive
+        except ApiError as e:
+            logger.warning("GPU-workload drain poll error: %s", e)
+        if remaining:
+            time.sleep(delay)
+            delay = min(delay * 2, poll_interval)
+    if remaining:
+        logger.warning("GPU workload pods still terminating at deadline: %s",
+                       sorted(remaining))
+    return ok
+
+
 def set_cc_state_label(k8s: K8sClient, node_name: str, state: str) -> bool:
     """Publish mode.state + derived ready.state (reference semantics,
     gpu_operator_eviction.py:262-295)."""

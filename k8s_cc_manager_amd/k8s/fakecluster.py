@@ -66,6 +66,8 @@ class FakeCluster:
         # (node, app) -> due time for pending operator actions
         self._pending_create: Dict[Tuple[str, str], float] = {}
         self._pending_delete: Dict[Tuple[str, str], float] = {}
+        self._pending_evict: Dict[Tuple[str, str, str], float] = {}
+        self._evictions: List[Tuple[str, str]] = []
         self._events: List[Dict[str, Any]] = []  # {"rv", "type", "node"}
 
         self._server: Optional[ThreadingHTTPServer] = None
@@ -113,8 +115,17 @@ class FakeCluster:
         with self._lock:
             return bool(self._nodes[name]["spec"].get("unschedulable"))
 
-    def add_pod(self, namespace: str, name: str, node: str, app: str) -> None:
+    def add_pod(self, namespace: str, name: str, node: str, app: str,
+                gpu_request: int = 0) -> None:
         with self._lock:
+            spec = {"nodeName": node}
+            if gpu_request:
+                spec["containers"] = [
+                    {
+                        "name": "main",
+                        "resources": {"requests": {"amd.com/gpu": str(gpu_request)}},
+                    }
+                ]
             self._pods[(namespace, name, node)] = {
                 "kind": "Pod",
                 "apiVersion": "v1",
@@ -123,7 +134,7 @@ class FakeCluster:
                     "namespace": namespace,
                     "labels": {"app": app},
                 },
-                "spec": {"nodeName": node},
+                "spec": spec,
                 "status": {"phase": "Running"},
             }
 
@@ -169,6 +180,10 @@ class FakeCluster:
         while not self._stopping:
             now = time.monotonic()
             with self._lock:
+                for key, due in list(self._pending_evict.items()):
+                    if now >= due:
+                        self._pods.pop(key, None)
+                        self._pending_evict.pop(key, None)
                 for node_name, node in self._nodes.items():
                     labels = node["metadata"]["labels"]
                     for comp_label, app in COMPONENT_APP_LABELS.items():
@@ -250,9 +265,42 @@ class FakeCluster:
                         and parts[4] == "pods"
                     ):
                         return self._list_pods(parts[3], qs)
+                    if parts == ["api", "v1", "pods"]:
+                        return self._list_pods(None, qs)
                     self._send_json(404, {"kind": "Status", "code": 404})
                 except (BrokenPipeError, ConnectionResetError):
                     pass
+
+            def do_POST(self) -> None:
+                # pods/eviction subresource: delete the pod after the
+                # configurable delete_delay (graceful termination)
+                url = urlparse(self.path)
+                parts = [p for p in url.path.split("/") if p]
+                length = int(self.headers.get("Content-Length", 0))
+                _ = self.rfile.read(length)
+                if (
+                    len(parts) == 7
+                    and parts[:3] == ["api", "v1", "namespaces"]
+                    and parts[4] == "pods"
+                    and parts[6] == "eviction"
+                ):
+                    ns, name = parts[3], parts[5]
+                    with cluster._lock:
+                        key = next(
+                            (k for k in cluster._pods if k[0] == ns and k[1] == name),
+                            None,
+                        )
+                        if key is None:
+                            return self._send_json(404, {"kind": "Status", "code": 404})
+                        cluster._evictions.append((ns, name))
+                        if cluster.delete_delay <= 0:
+                            cluster._pods.pop(key, None)
+                        else:
+                            cluster._pending_evict[key] = (
+                                time.monotonic() + cluster.delete_delay
+                            )
+                    return self._send_json(201, {"kind": "Status", "status": "Success"})
+                self._send_json(404, {"kind": "Status", "code": 404})
 
             def do_PATCH(self) -> None:
                 url = urlparse(self.path)
@@ -283,7 +331,7 @@ class FakeCluster:
                 with cluster._lock:
                     items = []
                     for (ns, _pname, node), pod in cluster._pods.items():
-                        if ns != namespace:
+                        if namespace is not None and ns != namespace:
                             continue
                         if want_node and node != want_node:
                             continue

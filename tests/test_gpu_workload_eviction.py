@@ -1,0 +1,99 @@
+"""GPU-workload pod eviction (Eviction API) around CC transitions."""
+
+import time
+
+from k8s_cc_manager_amd.core.manager import CCManager, ManagerConfig
+from k8s_cc_manager_amd.core.transition import TransitionEngine
+from k8s_cc_manager_amd.device.mock import MockBackend
+from k8s_cc_manager_amd.k8s.client import K8sClient
+from k8s_cc_manager_amd.k8s.eviction import (
+    COMPONENT_LABELS,
+    evict_gpu_workload_pods,
+)
+
+NODE = "node0"
+
+
+def test_evict_pod_subresource(fake_cluster):
+    cluster, url = fake_cluster
+    cluster.add_node(NODE)
+    cluster.add_pod("user-ns", "train-job", NODE, app="trainer", gpu_request=8)
+    k8s = K8sClient(url)
+    k8s.evict_pod("user-ns", "train-job")
+    deadline = time.monotonic() + 2
+    while time.monotonic() < deadline and cluster.pods_on(NODE):
+        time.sleep(0.02)
+    assert cluster.pods_on(NODE) == []
+    assert ("user-ns", "train-job") in cluster._evictions
+
+
+def test_evict_gpu_workload_pods_filters(fake_cluster):
+    """Only amd.com/gpu-requesting pods are evicted; CPU pods and
+    kube-system stay."""
+    cluster, url = fake_cluster
+    cluster.add_node(NODE)
+    cluster.add_pod("user-ns", "gpu-pod", NODE, app="t", gpu_request=1)
+    cluster.add_pod("user-ns", "cpu-pod", NODE, app="web", gpu_request=0)
+    cluster.add_pod("kube-system", "sys-gpu", NODE, app="sys", gpu_request=1)
+    k8s = K8sClient(url)
+    assert evict_gpu_workload_pods(k8s, NODE, timeout=5, poll_interval=0.02)
+    names = {p["metadata"]["name"] for p in cluster.pods_on(NODE)}
+    assert names == {"cpu-pod", "sys-gpu"}
+
+
+def test_transition_evicts_gpu_workloads_when_enabled(fake_cluster):
+    cluster, url = fake_cluster
+    cluster.add_node(NODE, labels={n: "true" for n in COMPONENT_LABELS})
+    cluster.add_pod("user-ns", "workload", NODE, app="train", gpu_request=4)
+    mgr = CCManager(
+        node_name=NODE,
+        default_mode="on",
+        host_cc=True,
+        k8s=K8sClient(url),
+        backend=MockBackend(num_gpus=2),
+        engine=TransitionEngine(),
+        config=ManagerConfig(
+            evict_components=True,
+            evict_gpu_workloads=True,
+            cordon_node=True,
+            eviction_timeout=5.0,
+            eviction_poll_interval=0.02,
+        ),
+    )
+    assert mgr.apply_mode("on")
+    assert ("user-ns", "workload") in cluster._evictions
+    assert all(
+        p["metadata"]["name"] != "workload" for p in cluster.pods_on(NODE)
+    )
+
+
+def test_workloads_untouched_by_default(fake_cluster):
+    cluster, url = fake_cluster
+    cluster.add_node(NODE, labels={n: "true" for n in COMPONENT_LABELS})
+    cluster.add_pod("user-ns", "workload", NODE, app="train", gpu_request=4)
+    mgr = CCManager(
+        node_name=NODE,
+        default_mode="on",
+        host_cc=True,
+        k8s=K8sClient(url),
+        backend=MockBackend(num_gpus=2),
+        engine=TransitionEngine(),
+        config=ManagerConfig(
+            evict_components=True,
+            cordon_node=False,
+            eviction_timeout=5.0,
+            eviction_poll_interval=0.02,
+        ),
+    )
+    assert mgr.apply_mode("on")
+    assert cluster._evictions == []
+    assert any(
+        p["metadata"]["name"] == "workload" for p in cluster.pods_on(NODE)
+    )
+
+
+def test_config_from_env(monkeypatch):
+    monkeypatch.setenv("EVICT_GPU_WORKLOADS", "true")
+    assert ManagerConfig.from_env().evict_gpu_workloads is True
+    monkeypatch.delenv("EVICT_GPU_WORKLOADS")
+    assert ManagerConfig.from_env().evict_gpu_workloads is False
