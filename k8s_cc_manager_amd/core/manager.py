@@ -118,6 +118,24 @@ class CCManager:
     def _set_state(self, state: str) -> None:
         eviction.set_cc_state_label(self.k8s, self.node_name, state)
 
+    _event_seq = 0
+
+    def _emit_event(self, reason: str, message: str, warning: bool = False) -> None:
+        """Best-effort core/v1 Event on the Node (kubectl describe node);
+        the reference has no event emission."""
+        CCManager._event_seq += 1
+        try:
+            self.k8s.create_event(
+                self.config.operator_namespace,
+                f"cc-{self.node_name}-{CCManager._event_seq}",
+                reason,
+                message,
+                self.node_name,
+                event_type="Warning" if warning else "Normal",
+            )
+        except Exception as e:
+            logger.debug("event emission failed: %s", e)
+
     def publish_capability_label(self) -> None:
         """Advertise whether this node can do GPU-CC at all:
         ``amd.com/gpu.cc.capable`` = host TEE support AND >=1 CC-capable
@@ -216,6 +234,7 @@ class CCManager:
     # transition wrappers
     # ------------------------------------------------------------------
     def _run_direct(self, mode: str, runner) -> bool:
+        self._emit_event("CCTransitionStarted", f"transitioning CC mode to {mode!r}")
         report: TransitionReport = runner()
         self.last_report = report
         METRICS.observe_transition(mode, report.ok, report.seconds, report.phases)
@@ -228,6 +247,18 @@ class CCManager:
             devices_changed=report.devices_changed,
             error=report.error,
         )
+        if report.ok:
+            self._emit_event(
+                "CCTransitionSucceeded",
+                f"CC mode {mode!r} applied in {report.seconds:.2f}s "
+                f"({len(report.devices_changed)} device(s) reset)",
+            )
+        else:
+            self._emit_event(
+                "CCTransitionFailed",
+                f"CC mode {mode!r} failed: {report.error}",
+                warning=True,
+            )
         self._set_state(mode if report.ok else STATE_FAILED)
         return report.ok
 

@@ -174,6 +174,43 @@ class K8sClient:
         _raise_for(resp)
         return resp.json()
 
+    def create_event(
+        self,
+        namespace: str,
+        name: str,
+        reason: str,
+        message: str,
+        node_name: str,
+        event_type: str = "Normal",
+    ) -> None:
+        """Post a core/v1 Event attached to the Node object (shows up in
+        ``kubectl describe node``)."""
+        import datetime
+
+        now = (
+            datetime.datetime.now(datetime.timezone.utc)
+            .strftime("%Y-%m-%dT%H:%M:%SZ")
+        )
+        body = {
+            "apiVersion": "v1",
+            "kind": "Event",
+            "metadata": {"name": name, "namespace": namespace},
+            "involvedObject": {"kind": "Node", "name": node_name, "apiVersion": "v1"},
+            "reason": reason,
+            "message": message,
+            "type": event_type,
+            "source": {"component": "amd-cc-manager", "host": node_name},
+            "firstTimestamp": now,
+            "lastTimestamp": now,
+            "count": 1,
+        }
+        resp = self._session.post(
+            f"{self.base_url}/api/v1/namespaces/{namespace}/events",
+            data=json.dumps(body),
+            headers={"Content-Type": "application/json"},
+        )
+        _raise_for(resp)
+
     def evict_pod(self, namespace: str, name: str) -> None:
         """Graceful eviction via the pods/eviction subresource (respects
         PodDisruptionBudgets, unlike a raw DELETE)."""

@@ -68,6 +68,7 @@ class FakeCluster:
         self._pending_delete: Dict[Tuple[str, str], float] = {}
         self._pending_evict: Dict[Tuple[str, str, str], float] = {}
         self._evictions: List[Tuple[str, str]] = []
+        self.k8s_events: List[Dict[str, Any]] = []
         self._events: List[Dict[str, Any]] = []  # {"rv", "type", "node"}
 
         self._server: Optional[ThreadingHTTPServer] = None
@@ -278,6 +279,18 @@ class FakeCluster:
                 parts = [p for p in url.path.split("/") if p]
                 length = int(self.headers.get("Content-Length", 0))
                 _ = self.rfile.read(length)
+                if (
+                    len(parts) == 5
+                    and parts[:3] == ["api", "v1", "namespaces"]
+                    and parts[4] == "events"
+                ):
+                    try:
+                        ev = json.loads(_ or b"{}")
+                    except Exception:
+                        ev = {}
+                    with cluster._lock:
+                        cluster.k8s_events.append(ev)
+                    return self._send_json(201, {"kind": "Status", "status": "Success"})
                 if (
                     len(parts) == 7
                     and parts[:3] == ["api", "v1", "namespaces"]
