@@ -95,7 +95,12 @@ def main(argv=None) -> int:
         return 1
 
     backend = get_backend(args.device_backend)
-    engine = TransitionEngine(attestor=_maybe_attestor(logger))
+    # the HIP attestor resolves devices by PCI bdf — only meaningful for
+    # real backends (mock bdfs have no HIP device behind them)
+    from .device.mock import MockBackend
+
+    attestor = None if isinstance(backend, MockBackend) else _maybe_attestor(logger)
+    engine = TransitionEngine(attestor=attestor)
     manager = CCManager(
         node_name=args.node_name,
         default_mode=default_mode,
