@@ -25,10 +25,37 @@ def bind(path):
     return fn
 
 
+VARIANT_FLAGS = {
+    "base": [],
+    "pf": ["-DCC_EXP_PREFETCH_FIRST=1"],
+    "lgkm": ["-DCC_EXP_LGKM=1"],
+    "both": ["-DCC_EXP_PREFETCH_FIRST=1", "-DCC_EXP_LGKM=1"],
+}
+
+
+def build_variants():
+    """Build the experiment .so set if missing (they are gitignored)."""
+    import subprocess
+
+    src = Path(__file__).resolve().parent.parent / "k8s_cc_manager_amd/ops/attest_kernels.hip"
+    outdir = Path("expbuild")
+    outdir.mkdir(exist_ok=True)
+    for name, flags in VARIANT_FLAGS.items():
+        out = outdir / f"libcc_{name}.so"
+        if out.exists() and out.stat().st_mtime >= src.stat().st_mtime:
+            continue
+        subprocess.run(
+            ["hipcc", "--offload-arch=gfx950", "-O3", "-std=c++17", "-shared",
+             "-fPIC", "-Wno-unused-value", *flags, str(src), "-o", str(out)],
+            check=True,
+        )
+
+
 def main():
     n = int(sys.argv[1]) if len(sys.argv) > 1 else 8192
     rounds = int(sys.argv[2]) if len(sys.argv) > 2 else 6
     iters = 10
+    build_variants()
     variants = {}
     for so in sorted(Path("expbuild").glob("libcc_*.so")):
         variants[so.stem.replace("libcc_", "")] = bind(str(so))
