@@ -1,0 +1,135 @@
+#!/usr/bin/env python3
+"""Watch-loop soak: continuous label flips against the live manager.
+
+Runs the full manager (watch loop included) against the in-process fake
+API server with the shadow GPU backend + real attestation, flipping the
+desired mode every ``--period`` seconds for ``--duration`` seconds.
+Asserts at the end: every transition succeeded, the applied state always
+caught up, and process RSS / VRAM stayed flat. Prints one JSON line.
+"""
+
+import argparse
+import json
+import sys
+import threading
+import time
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+
+
+def rss_mb() -> float:
+    import resource
+
+    return resource.getrusage(resource.RUSAGE_SELF).ru_maxrss / 1024.0
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--duration", type=float, default=360.0)
+    ap.add_argument("--period", type=float, default=0.3)
+    ap.add_argument("--attest-dim", type=int, default=512)
+    ap.add_argument("--mock", action="store_true")
+    args = ap.parse_args()
+
+    import os
+
+    os.environ.setdefault("CC_EVENT_LOG", "/tmp/soak_events.jsonl")
+    Path(os.environ["CC_EVENT_LOG"]).unlink(missing_ok=True)
+
+    from k8s_cc_manager_amd.core.manager import CCManager, ManagerConfig
+    from k8s_cc_manager_amd.core.transition import TransitionEngine
+    from k8s_cc_manager_amd.k8s.client import K8sClient
+    from k8s_cc_manager_amd.k8s.eviction import COMPONENT_LABELS
+    from k8s_cc_manager_amd.k8s.fakecluster import FakeCluster
+    from k8s_cc_manager_amd.labels import CC_MODE_LABEL, CC_STATE_LABEL
+    from k8s_cc_manager_amd.utils import eventlog
+
+    use_gpu = False
+    attestor = None
+    if not args.mock:
+        try:
+            import torch
+
+            use_gpu = torch.cuda.is_available()
+        except Exception:
+            pass
+    if use_gpu:
+        from k8s_cc_manager_amd.device.shadow import ShadowBackend
+        from k8s_cc_manager_amd.ops import attest
+
+        backend = ShadowBackend(device_indices=[0])
+        attestor = lambda d: attest.attest_device(  # noqa: E731
+            d.hip_index, gemm_dim=args.attest_dim
+        )
+    else:
+        from k8s_cc_manager_amd.device.mock import MockBackend
+
+        backend = MockBackend(num_gpus=2)
+
+    cluster = FakeCluster(operator_tick=0.01)
+    url = cluster.start()
+    cluster.add_node("soak", labels={n: "true" for n in COMPONENT_LABELS})
+    mgr = CCManager(
+        node_name="soak",
+        default_mode="off",
+        host_cc=True,
+        k8s=K8sClient(url),
+        backend=backend,
+        engine=TransitionEngine(attestor=attestor),
+        config=ManagerConfig(
+            evict_components=True,
+            cordon_node=True,
+            eviction_timeout=20.0,
+            eviction_poll_interval=0.01,
+            watch_timeout_seconds=5,
+            reconnect_backoff=0.1,
+            readiness_file="/tmp/.soak-ready",
+        ),
+    )
+    t = threading.Thread(target=mgr.run, daemon=True)
+    t.start()
+    time.sleep(2.0)
+
+    rss0 = rss_mb()
+    modes = ["on", "off", "devtools"]
+    flips = 0
+    t_end = time.monotonic() + args.duration
+    while time.monotonic() < t_end:
+        mode = modes[flips % len(modes)]
+        cluster.set_node_label("soak", CC_MODE_LABEL, mode)
+        flips += 1
+        time.sleep(args.period)
+    # let the last transition settle
+    deadline = time.monotonic() + 30
+    final_mode = modes[(flips - 1) % len(modes)]
+    while time.monotonic() < deadline:
+        if cluster.node_labels("soak").get(CC_STATE_LABEL) == final_mode:
+            break
+        time.sleep(0.1)
+    mgr.stop_event.set()
+    t.join(timeout=15)
+
+    events = eventlog.read_transitions(Path(os.environ["CC_EVENT_LOG"]))
+    failures = [e for e in events if not e["ok"]]
+    settled = cluster.node_labels("soak").get(CC_STATE_LABEL) == final_mode
+    result = {
+        "soak_seconds": args.duration,
+        "label_flips": flips,
+        "transitions_applied": len(events),
+        "transition_failures": len(failures),
+        "settled_on_final_mode": settled,
+        "rss_growth_mb": round(rss_mb() - rss0, 1),
+        "mean_transition_s": round(
+            sum(e["seconds"] for e in events) / max(len(events), 1), 4
+        ),
+        "tier": "shadow+hip-attest" if use_gpu else "mock",
+        "ok": bool(not failures and settled and len(events) > 0),
+    }
+    print(json.dumps(result))
+    cluster.stop()
+    return 0 if result["ok"] else 1
+
+
+if __name__ == "__main__":
+    sys.exit(main())
