@@ -5,17 +5,21 @@
 // register (/root/reference/main.py:523-529). Here the device must
 // additionally PROVE it executes correctly inside the TEE:
 //
-//   1. mfma_gemm_bf16   — LDS-tiled bf16 GEMM on the MFMA matrix cores
-//                         (v_mfma_f32_16x16x32_bf16), exercising HBM ->
-//                         LDS -> VGPR -> MFMA -> HBM end to end;
-//   2. ref_gemm_f32     — plain VALU fp32 GEMM of the same inputs: the
+//   1. bf16 MFMA GEMMs  — mfma_gemm_bf16 (128x128 step-3 structure),
+//                         mfma_gemm_bf16_256 / _256w (256x256 8-phase
+//                         deep-pipelined, 16x16x32 / 32x32x16 shapes) —
+//                         HBM -> LDS-DMA -> VGPR -> MFMA -> HBM;
+//   2. fp8 MFMA GEMMs   — mfma_gemm_fp8_128 / _256: MX-scaled OCP e4m3
+//                         (v_mfma_scale_*_f8f6f4, unit e8m0 scales);
+//   3. ref_gemm_f32     — plain VALU fp32 GEMM of the same inputs: the
 //                         independent on-device ground truth (inputs are
-//                         small integers, so both paths must agree
-//                         BITWISE — any mismatch is silicon/TEE trouble);
-//   3. lds_probe        — LDS cell sweep with rotating patterns;
-//   4. hbm_probe        — vectorized streaming copy + checksum (HBM3E
+//                         small integers, so the bf16 AND fp8 MFMA paths
+//                         must both agree BITWISE — any mismatch is
+//                         silicon/TEE trouble);
+//   4. lds_probe        — LDS cell sweep with rotating patterns;
+//   5. hbm_probe        — vectorized streaming copy + checksum (HBM3E
 //                         path, reported as GB/s);
-//   5. xGMI peer visibility via hipDeviceCanAccessPeer.
+//   6. xGMI peer visibility via hipDeviceCanAccessPeer.
 //
 // Layout notes (from the CDNA4 guides): wave = 64 lanes; for
 // v_mfma_f32_16x16x32_bf16 each lane carries 8 bf16 of A and B and 4
