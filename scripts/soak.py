@@ -19,9 +19,12 @@ sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
 
 
 def rss_mb() -> float:
-    import resource
-
-    return resource.getrusage(resource.RUSAGE_SELF).ru_maxrss / 1024.0
+    """Current RSS (not peak): /proc/self/status VmRSS."""
+    with open("/proc/self/status") as f:
+        for line in f:
+            if line.startswith("VmRSS:"):
+                return int(line.split()[1]) / 1024.0
+    return 0.0
 
 
 def main():
@@ -92,13 +95,17 @@ def main():
     time.sleep(2.0)
 
     rss0 = rss_mb()
+    rss_mid = rss0
     modes = ["on", "off", "devtools"]
     flips = 0
-    t_end = time.monotonic() + args.duration
+    t_start = time.monotonic()
+    t_end = t_start + args.duration
     while time.monotonic() < t_end:
         mode = modes[flips % len(modes)]
         cluster.set_node_label("soak", CC_MODE_LABEL, mode)
         flips += 1
+        if rss_mid == rss0 and time.monotonic() - t_start > args.duration / 2:
+            rss_mid = rss_mb()
         time.sleep(args.period)
     # let the last transition settle
     deadline = time.monotonic() + 30
@@ -120,6 +127,7 @@ def main():
         "transition_failures": len(failures),
         "settled_on_final_mode": settled,
         "rss_growth_mb": round(rss_mb() - rss0, 1),
+        "rss_second_half_growth_mb": round(rss_mb() - rss_mid, 1),
         "mean_transition_s": round(
             sum(e["seconds"] for e in events) / max(len(events), 1), 4
         ),
