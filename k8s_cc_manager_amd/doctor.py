@@ -55,11 +55,32 @@ def collect(run_attest: bool = False, gemm_dim: int = 512) -> Dict[str, Any]:
         be = AmdSmiBackend()
         devices, count = be.find_devices()
         smi["available"] = True
-        smi["devices"] = [
-            {"bdf": d.bdf, "name": d.name, "cc_mode": d.query_cc_mode(),
-             "fabric_mode": d.query_fabric_mode()}
-            for d in devices
-        ]
+        smi["devices"] = []
+        for d in devices:
+            entry = {
+                "bdf": d.bdf,
+                "name": d.name,
+                "cc_mode": d.query_cc_mode(),
+                "fabric_mode": d.query_fabric_mode(),
+            }
+            # best-effort telemetry (fields vary by stack version)
+            try:
+                import amdsmi
+
+                vram = amdsmi.amdsmi_get_gpu_vram_info(d._handle)
+                entry["vram_total_mb"] = int(vram.get("vram_total", 0)) // (1 << 20) \
+                    if isinstance(vram, dict) else None
+            except Exception:
+                pass
+            try:
+                import amdsmi
+
+                entry["power_w"] = amdsmi.amdsmi_get_power_info(d._handle).get(
+                    "average_socket_power"
+                )
+            except Exception:
+                pass
+            smi["devices"].append(entry)
         smi["count"] = count
     except Exception as e:
         smi["error"] = str(e)
