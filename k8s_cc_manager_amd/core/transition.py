@@ -42,10 +42,6 @@ logger = logging.getLogger(__name__)
 Attestor = Callable[[CCDevice], None]
 
 
-class TransitionError(Exception):
-    """A transition failed; the node must be labeled 'failed'."""
-
-
 @dataclass
 class TransitionReport:
     ok: bool

@@ -72,3 +72,32 @@ def test_concurrency_actually_parallel():
 
     out = DeviceExecutor().run("p", devices, fn)
     assert len(out) == 8
+
+
+def test_in_process_fabric_barrier_synchronizes_two_engines():
+    """Two in-process hive managers sharing one FabricBarrier cannot
+    cross the stage->reset seam independently."""
+    import time
+
+    from k8s_cc_manager_amd.core.transition import TransitionEngine
+    from k8s_cc_manager_amd.parallel.fabric import FabricBarrier
+
+    barrier = FabricBarrier(parties=2)
+    backends = [MockBackend(num_gpus=1), MockBackend(num_gpus=1)]
+    engines = [TransitionEngine(barrier=barrier) for _ in backends]
+    results = [None, None]
+
+    def run(i):
+        devices, _ = backends[i].find_devices()
+        results[i] = engines[i].apply_fabric_mode(devices)
+
+    t0 = threading.Thread(target=run, args=(0,))
+    t1 = threading.Thread(target=run, args=(1,))
+    t0.start()
+    time.sleep(0.05)  # stagger: the barrier must hold engine 0 back
+    t1.start()
+    t0.join(timeout=30)
+    t1.join(timeout=30)
+    assert results[0].ok and results[1].ok
+    assert backends[0].device(0).query_fabric_mode() == "on"
+    assert backends[1].device(0).query_fabric_mode() == "on"
