@@ -13,6 +13,14 @@ one-process-per-GPU), the stage/reset seam needs an explicit barrier:
   ``torch.distributed`` (gloo on CPU, RCCL on ROCm), used when each
   rank owns one GPU of the hive. torch is imported lazily so the
   control-plane daemon itself never depends on it.
+
+INVARIANT: every participant must take the same phase path through the
+engine — the conditional phase-1 (force fabric off) barrier only fires
+for participants whose devices were fabric-on, so a MIXED initial state
+across participants would mismatch barrier counts and deadlock the
+hive. Hive-wide transitions therefore require a symmetric starting
+state (which the reconcile guarantees: one manager drives one node's
+hive, and the bench toggles all ranks' devices in lockstep).
 """
 
 from __future__ import annotations
