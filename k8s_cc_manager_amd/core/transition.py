@@ -106,7 +106,11 @@ class TransitionEngine:
                 self.executor.run(
                     "fabric-stage-off", fabric_on, lambda d: d.set_fabric_mode(FABRIC_OFF)
                 )
-                self._barrier_wait()
+            # unconditional: every hive participant arrives at the
+            # phase-1 seam even with nothing to reset, so mixed initial
+            # states cannot mismatch barrier counts
+            self._barrier_wait()
+            if fabric_on:
                 self.executor.run("fabric-reset", fabric_on, lambda d: d.reset())
                 self.executor.run(
                     "fabric-verify",
@@ -181,7 +185,8 @@ class TransitionEngine:
                 self.executor.run(
                     "fabric-off-stage", stuck_on, lambda d: d.set_fabric_mode(FABRIC_OFF)
                 )
-                self._barrier_wait()
+            self._barrier_wait()  # unconditional (see apply_cc_mode)
+            if stuck_on:
                 self.executor.run("fabric-off-reset", stuck_on, lambda d: d.reset())
                 self.executor.run(
                     "fabric-off-verify",

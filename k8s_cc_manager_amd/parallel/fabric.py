@@ -14,13 +14,10 @@ one-process-per-GPU), the stage/reset seam needs an explicit barrier:
   rank owns one GPU of the hive. torch is imported lazily so the
   control-plane daemon itself never depends on it.
 
-INVARIANT: every participant must take the same phase path through the
-engine — the conditional phase-1 (force fabric off) barrier only fires
-for participants whose devices were fabric-on, so a MIXED initial state
-across participants would mismatch barrier counts and deadlock the
-hive. Hive-wide transitions therefore require a symmetric starting
-state (which the reconcile guarantees: one manager drives one node's
-hive, and the bench toggles all ranks' devices in lockstep).
+Barrier counts are matched by construction: the engine's phase-1 and
+stage->reset seams call wait() UNCONDITIONALLY for every participant
+(even one with no devices to reset), so mixed initial states across
+participants cannot deadlock the hive.
 """
 
 from __future__ import annotations

@@ -101,3 +101,33 @@ def test_in_process_fabric_barrier_synchronizes_two_engines():
     assert results[0].ok and results[1].ok
     assert backends[0].device(0).query_fabric_mode() == "on"
     assert backends[1].device(0).query_fabric_mode() == "on"
+
+
+def test_fabric_barrier_mixed_initial_state_no_deadlock():
+    """One participant's hive is already fabric-on, the other's is off:
+    barrier counts still match (unconditional seam waits)."""
+    import time
+
+    from k8s_cc_manager_amd.core.transition import TransitionEngine
+    from k8s_cc_manager_amd.parallel.fabric import FabricBarrier
+
+    barrier = FabricBarrier(parties=2)
+    be_on = MockBackend(num_gpus=1, initial_fabric_mode="on")
+    be_off = MockBackend(num_gpus=1)
+    backends = [be_on, be_off]
+    engines = [TransitionEngine(barrier=barrier) for _ in backends]
+    results = [None, None]
+
+    def run(i):
+        devices, _ = backends[i].find_devices()
+        results[i] = engines[i].apply_fabric_mode(devices)
+
+    threads = [threading.Thread(target=run, args=(i,)) for i in range(2)]
+    t0 = time.monotonic()
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=30)
+    assert time.monotonic() - t0 < 25, "barrier deadlock"
+    assert results[0].ok and results[1].ok
+    assert all(b.device(0).query_fabric_mode() == "on" for b in backends)
