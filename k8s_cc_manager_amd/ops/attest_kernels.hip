@@ -1101,9 +1101,17 @@ static int launch_mfma_gemm(const void* A, const void* Bt, void* C, int M,
     // enable the XCD remap only past the 256 MiB Infinity Cache
     long ws = 2L * K * (M + N) + 4L * M * N;
     int swz_on = ws > (256L << 20) ? 1 : 0;
-    hipLaunchKernelGGL(mfma_gemm_bf16_256, grid, dim3(512), 0, 0,
-                       (const bf16*)A, (const bf16*)Bt, (float*)C, M, N, K,
-                       swz_on);
+    // measured across boxes: the 32x32x16 shape wins at <=1 block/CU
+    // grids (935 vs 848 TF @4096^3), the 16x16x32 shape past that
+    // (1129 vs 1096 @8192^3) — profiles/gemm_final_sweep_2026-09-13.json
+    if ((long)grid.x * grid.y <= 256)
+      hipLaunchKernelGGL(mfma_gemm_bf16_256w, grid, dim3(512), 0, 0,
+                         (const bf16*)A, (const bf16*)Bt, (float*)C, M, N, K,
+                         swz_on);
+    else
+      hipLaunchKernelGGL(mfma_gemm_bf16_256, grid, dim3(512), 0, 0,
+                         (const bf16*)A, (const bf16*)Bt, (float*)C, M, N, K,
+                         swz_on);
     return 0;
   }
   if (M % BM || N % BN || K % BK) return -2;
