@@ -960,9 +960,10 @@ constexpr int kMaxDevices = 64;
 
 struct ProbeCtx {
   int dim = 0;
+  int ref_valid = 0;  // the fp32 reference of the deterministic fill
   bf16 *dA = nullptr, *dB = nullptr;
   unsigned char *dA8 = nullptr, *dB8 = nullptr;
-  float *dC = nullptr, *dRef = nullptr, *dErr = nullptr;
+  float *dC = nullptr, *dRef = nullptr, *dHbm = nullptr, *dErr = nullptr;
   unsigned long long* dSum = nullptr;
   uint32_t* dFail = nullptr;
   int* dLive = nullptr;
@@ -979,6 +980,7 @@ static void ctx_release(ProbeCtx& c) {
   if (c.dB8) (void)hipFree(c.dB8);
   if (c.dC) (void)hipFree(c.dC);
   if (c.dRef) (void)hipFree(c.dRef);
+  if (c.dHbm) (void)hipFree(c.dHbm);
   if (c.dErr) (void)hipFree(c.dErr);
   if (c.dSum) (void)hipFree(c.dSum);
   if (c.dFail) (void)hipFree(c.dFail);
@@ -1003,6 +1005,7 @@ static hipError_t ctx_acquire(int device, int dim, ProbeCtx** out) {
   if ((e = hipMalloc(&c.dB8, elems)) != hipSuccess) return e;
   if ((e = hipMalloc(&c.dC, elems * sizeof(float))) != hipSuccess) return e;
   if ((e = hipMalloc(&c.dRef, elems * sizeof(float))) != hipSuccess) return e;
+  if ((e = hipMalloc(&c.dHbm, elems * sizeof(float))) != hipSuccess) return e;
   if ((e = hipMalloc(&c.dErr, sizeof(float))) != hipSuccess) return e;
   if ((e = hipMalloc(&c.dSum, sizeof(unsigned long long))) != hipSuccess) return e;
   if ((e = hipMalloc(&c.dFail, sizeof(uint32_t))) != hipSuccess) return e;
@@ -1239,14 +1242,24 @@ int cc_attest_device(int device, int gemm_dim, struct CcAttestReport* rep) {
   rep->gemm_tflops = 2.0 * D * (double)D * D / (rep->gemm_ms * 1e-3) / 1e12;
 
   // -- VALU reference + compare ---------------------------------------
-  CC_CHECK(hipEventRecord(ev0, 0));
-  {
-    dim3 rblock(16, 16), rgrid((D + 15) / 16, (D + 15) / 16);
-    hipLaunchKernelGGL(ref_gemm_f32, rgrid, rblock, 0, 0, dA, dB, dRef, D, D, D);
+  // The fill is deterministic per dim, so the fp32 reference is a
+  // constant: computed once per context (the dominant probe cost —
+  // ~1.9 ms of a 2.4 ms warm probe at dim 1024 — disappears from the
+  // steady-state transition path).
+  if (!ctx->ref_valid) {
+    CC_CHECK(hipEventRecord(ev0, 0));
+    {
+      dim3 rblock(16, 16), rgrid((D + 15) / 16, (D + 15) / 16);
+      hipLaunchKernelGGL(ref_gemm_f32, rgrid, rblock, 0, 0, dA, dB, dRef, D, D,
+                         D);
+    }
+    CC_CHECK(hipEventRecord(ev1, 0));
+    CC_CHECK(hipEventSynchronize(ev1));
+    rep->ref_ms = event_ms(ev0, ev1);
+    ctx->ref_valid = 1;
+  } else {
+    rep->ref_ms = 0.0;  // cached
   }
-  CC_CHECK(hipEventRecord(ev1, 0));
-  CC_CHECK(hipEventSynchronize(ev1));
-  rep->ref_ms = event_ms(ev0, ev1);
 
   hipLaunchKernelGGL(max_abs_diff, dim3(1024), dim3(256), 0, 0, dC, dRef,
                      elems, dErr);
@@ -1302,11 +1315,11 @@ int cc_attest_device(int device, int gemm_dim, struct CcAttestReport* rep) {
   CC_CHECK(hipMemcpy(&rep->lds_failures, dFail, sizeof(uint32_t),
                      hipMemcpyDeviceToHost));
 
-  // -- HBM probe: reuse C/Ref buffers as src/dst -----------------------
+  // -- HBM probe (dedicated dst: dRef is a cached constant now) --------
   long n4 = elems / 4;
   CC_CHECK(hipEventRecord(ev0, 0));
   hipLaunchKernelGGL(hbm_copy_f4, dim3(4096), dim3(256), 0, 0,
-                     (const float4v*)dC, (float4v*)dRef, n4);
+                     (const float4v*)dC, (float4v*)ctx->dHbm, n4);
   CC_CHECK(hipEventRecord(ev1, 0));
   CC_CHECK(hipEventSynchronize(ev1));
   rep->hbm_ms = event_ms(ev0, ev1);
