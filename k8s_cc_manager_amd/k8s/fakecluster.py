@@ -389,6 +389,24 @@ class FakeCluster:
                 try:
                     while time.monotonic() < deadline:
                         with cluster._lock:
+                            if last_sent < cluster._compacted_rv:
+                                # our cursor fell off the trimmed event
+                                # log: real apiservers signal this with
+                                # a 410 so the client re-lists instead
+                                # of silently missing events
+                                send_chunk(
+                                    {
+                                        "type": "ERROR",
+                                        "object": {
+                                            "kind": "Status",
+                                            "code": 410,
+                                            "reason": "Expired",
+                                            "message": f"too old resource version: {last_sent}",
+                                        },
+                                    }
+                                )
+                                self.wfile.write(b"0\r\n\r\n")
+                                return
                             pending = [
                                 e
                                 for e in cluster._events

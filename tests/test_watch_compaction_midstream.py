@@ -1,0 +1,81 @@
+"""A watch whose cursor is trimmed away MID-STREAM must get a 410 ERROR
+event (not silently miss events) and the manager must resync."""
+
+import threading
+import time
+
+from k8s_cc_manager_amd.core.manager import CCManager, ManagerConfig
+from k8s_cc_manager_amd.core.transition import TransitionEngine
+from k8s_cc_manager_amd.device.mock import MockBackend
+from k8s_cc_manager_amd.k8s.client import K8sClient
+from k8s_cc_manager_amd.labels import CC_MODE_LABEL, CC_STATE_LABEL
+
+NODE = "node0"
+
+
+def test_midstream_compaction_emits_410(fake_cluster):
+    cluster, url = fake_cluster
+    cluster.add_node(NODE)
+    k8s = K8sClient(url)
+    rv = k8s.get_node(NODE)["metadata"]["resourceVersion"]
+
+    seen = []
+
+    def watcher():
+        for event in k8s.watch_node(NODE, resource_version=rv, timeout_seconds=5):
+            seen.append(event)
+            if event.get("type") == "ERROR":
+                return
+
+    t = threading.Thread(target=watcher)
+    t.start()
+    time.sleep(0.2)
+    # burst enough churn to trim the event log past the watcher's cursor
+    for i in range(5000):
+        cluster.set_node_label(NODE, "churn", str(i))
+    t.join(timeout=20)
+    assert any(
+        e.get("type") == "ERROR" and e["object"].get("code") == 410 for e in seen
+    ), f"no 410 in {len(seen)} events"
+
+
+def test_manager_survives_midstream_compaction(fake_cluster):
+    cluster, url = fake_cluster
+    cluster.add_node(NODE, labels={CC_MODE_LABEL: "off"})
+    backend = MockBackend(num_gpus=1)
+    mgr = CCManager(
+        node_name=NODE,
+        default_mode="off",
+        host_cc=True,
+        k8s=K8sClient(url),
+        backend=backend,
+        engine=TransitionEngine(),
+        config=ManagerConfig(
+            evict_components=False,
+            cordon_node=False,
+            watch_timeout_seconds=5,
+            reconnect_backoff=0.05,
+            readiness_file="/tmp/.cc-compact-test",
+        ),
+    )
+    t = threading.Thread(target=mgr.run, daemon=True)
+    t.start()
+    deadline = time.monotonic() + 5
+    while time.monotonic() < deadline:
+        if cluster.node_labels(NODE).get(CC_STATE_LABEL) == "off":
+            break
+        time.sleep(0.02)
+
+    # churn past the trim boundary, then flip the mode: the 410-resync
+    # path must pick up the new desired mode
+    for i in range(5000):
+        cluster.set_node_label(NODE, "churn", str(i))
+    cluster.set_node_label(NODE, CC_MODE_LABEL, "devtools")
+    deadline = time.monotonic() + 15
+    while time.monotonic() < deadline:
+        if cluster.node_labels(NODE).get(CC_STATE_LABEL) == "devtools":
+            break
+        time.sleep(0.05)
+    assert cluster.node_labels(NODE).get(CC_STATE_LABEL) == "devtools"
+    mgr.stop_event.set()
+    t.join(timeout=10)
