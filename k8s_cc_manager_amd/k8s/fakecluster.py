@@ -52,11 +52,13 @@ class FakeCluster:
         schedule_delay: float = 0.0,
         delete_delay: float = 0.0,
         operator_tick: float = 0.02,
+        event_log_max: int = 4096,
     ):
         self.operator_namespace = operator_namespace
         self.schedule_delay = schedule_delay
         self.delete_delay = delete_delay
         self._operator_tick = operator_tick
+        self._event_log_max = event_log_max
 
         self._lock = threading.Condition()
         self._rv = 0
@@ -169,8 +171,8 @@ class FakeCluster:
                 "node": json.loads(json.dumps(self._nodes[name])),
             }
         )
-        if len(self._events) > 4096:
-            self._events = self._events[-2048:]
+        if len(self._events) > self._event_log_max:
+            self._events = self._events[-(self._event_log_max // 2):]
             self._compacted_rv = max(self._compacted_rv, self._events[0]["rv"] - 1)
         self._lock.notify_all()
 

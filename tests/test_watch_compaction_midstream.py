@@ -13,8 +13,11 @@ from k8s_cc_manager_amd.labels import CC_MODE_LABEL, CC_STATE_LABEL
 NODE = "node0"
 
 
-def test_midstream_compaction_emits_410(fake_cluster):
-    cluster, url = fake_cluster
+def test_midstream_compaction_emits_410():
+    from k8s_cc_manager_amd.k8s.fakecluster import FakeCluster
+
+    cluster = FakeCluster(event_log_max=100)
+    url = cluster.start()
     cluster.add_node(NODE)
     k8s = K8sClient(url)
     rv = k8s.get_node(NODE)["metadata"]["resourceVersion"]
@@ -22,25 +25,32 @@ def test_midstream_compaction_emits_410(fake_cluster):
     seen = []
 
     def watcher():
-        for event in k8s.watch_node(NODE, resource_version=rv, timeout_seconds=5):
+        # deliberately SLOW consumer: backpressure lets the server-side
+        # event log trim past this watcher's cursor
+        for event in k8s.watch_node(NODE, resource_version=rv, timeout_seconds=15):
             seen.append(event)
             if event.get("type") == "ERROR":
                 return
+            time.sleep(0.05)
 
     t = threading.Thread(target=watcher)
     t.start()
     time.sleep(0.2)
     # burst enough churn to trim the event log past the watcher's cursor
-    for i in range(5000):
+    for i in range(1500):
         cluster.set_node_label(NODE, "churn", str(i))
-    t.join(timeout=20)
+    t.join(timeout=30)
+    cluster.stop()
     assert any(
         e.get("type") == "ERROR" and e["object"].get("code") == 410 for e in seen
     ), f"no 410 in {len(seen)} events"
 
 
-def test_manager_survives_midstream_compaction(fake_cluster):
-    cluster, url = fake_cluster
+def test_manager_survives_midstream_compaction():
+    from k8s_cc_manager_amd.k8s.fakecluster import FakeCluster
+
+    cluster = FakeCluster(event_log_max=200)
+    url = cluster.start()
     cluster.add_node(NODE, labels={CC_MODE_LABEL: "off"})
     backend = MockBackend(num_gpus=1)
     mgr = CCManager(
@@ -68,7 +78,7 @@ def test_manager_survives_midstream_compaction(fake_cluster):
 
     # churn past the trim boundary, then flip the mode: the 410-resync
     # path must pick up the new desired mode
-    for i in range(5000):
+    for i in range(400):
         cluster.set_node_label(NODE, "churn", str(i))
     cluster.set_node_label(NODE, CC_MODE_LABEL, "devtools")
     deadline = time.monotonic() + 15
@@ -79,3 +89,4 @@ def test_manager_survives_midstream_compaction(fake_cluster):
     assert cluster.node_labels(NODE).get(CC_STATE_LABEL) == "devtools"
     mgr.stop_event.set()
     t.join(timeout=10)
+    cluster.stop()
