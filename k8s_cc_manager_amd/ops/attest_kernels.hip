@@ -954,7 +954,8 @@ __global__ void hbm_copy_f4(const float4v* __restrict__ src,
 // daemon that attests after every transition: allocations, events and
 // the liveness scratch word are cached so steady-state probes pay only
 // kernel time (first call measured ~225 ms of hipMalloc/teardown
-// overhead at dim=1024; cached, the probe is ~2 ms).
+// overhead at dim=1024; with the cached context AND the cached fp32
+// reference below, the warm probe is ~0.4 ms).
 // ---------------------------------------------------------------------------
 constexpr int kMaxDevices = 64;
 
