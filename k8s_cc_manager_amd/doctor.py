@@ -68,8 +68,10 @@ def collect(run_attest: bool = False, gemm_dim: int = 512) -> Dict[str, Any]:
                 import amdsmi
 
                 vram = amdsmi.amdsmi_get_gpu_vram_info(d._handle)
-                entry["vram_total_mb"] = int(vram.get("vram_total", 0)) // (1 << 20) \
-                    if isinstance(vram, dict) else None
+                if isinstance(vram, dict):
+                    raw = int(vram.get("vram_total", 0))
+                    # stacks disagree on units: treat small values as MB
+                    entry["vram_total_mb"] = raw if raw < (1 << 24) else raw >> 20
             except Exception:
                 pass
             try:
