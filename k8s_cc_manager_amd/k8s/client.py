@@ -61,12 +61,20 @@ class K8sClient:
     ):
         self.base_url = base_url.rstrip("/")
         self._session = session or requests.Session()
-        self._session.verify = verify
-        if cert:
-            self._session.cert = cert
+        # requests 2.33 does NOT honor a CA-bundle path set on
+        # Session.verify (only per-request verify works) — found by the
+        # TLS test; pass it on every request instead.
+        self._verify = verify
+        self._cert = cert
         if token:
             self._session.headers["Authorization"] = f"Bearer {token}"
         self._session.headers["Accept"] = "application/json"
+
+    def _request(self, method: str, url: str, **kwargs) -> requests.Response:
+        kwargs.setdefault("verify", self._verify)
+        if self._cert:
+            kwargs.setdefault("cert", self._cert)
+        return self._session.request(method, url, **kwargs)
 
     # -- construction ---------------------------------------------------
     @classmethod
@@ -127,14 +135,15 @@ class K8sClient:
 
     # -- nodes ----------------------------------------------------------
     def get_node(self, name: str) -> Dict[str, Any]:
-        resp = self._session.get(f"{self.base_url}/api/v1/nodes/{name}")
+        resp = self._request("GET", f"{self.base_url}/api/v1/nodes/{name}")
         _raise_for(resp)
         return resp.json()
 
     def patch_node_labels(self, name: str, labels: Dict[str, Optional[str]]) -> Dict[str, Any]:
         """Merge-patch only the given labels (None deletes a label)."""
         patch = {"metadata": {"labels": labels}}
-        resp = self._session.patch(
+        resp = self._request(
+            "PATCH",
             f"{self.base_url}/api/v1/nodes/{name}",
             data=json.dumps(patch),
             headers={"Content-Type": "application/strategic-merge-patch+json"},
@@ -145,7 +154,8 @@ class K8sClient:
     def set_node_unschedulable(self, name: str, unschedulable: bool) -> Dict[str, Any]:
         """Cordon (True) / uncordon (False) the node."""
         patch = {"spec": {"unschedulable": unschedulable or None}}
-        resp = self._session.patch(
+        resp = self._request(
+            "PATCH",
             f"{self.base_url}/api/v1/nodes/{name}",
             data=json.dumps(patch),
             headers={"Content-Type": "application/strategic-merge-patch+json"},
@@ -170,7 +180,7 @@ class K8sClient:
             if namespace
             else f"{self.base_url}/api/v1/pods"  # all namespaces
         )
-        resp = self._session.get(url, params=params)
+        resp = self._request("GET", url, params=params)
         _raise_for(resp)
         return resp.json()
 
@@ -204,7 +214,8 @@ class K8sClient:
             "lastTimestamp": now,
             "count": 1,
         }
-        resp = self._session.post(
+        resp = self._request(
+            "POST",
             f"{self.base_url}/api/v1/namespaces/{namespace}/events",
             data=json.dumps(body),
             headers={"Content-Type": "application/json"},
@@ -219,7 +230,8 @@ class K8sClient:
             "kind": "Eviction",
             "metadata": {"name": name, "namespace": namespace},
         }
-        resp = self._session.post(
+        resp = self._request(
+            "POST",
             f"{self.base_url}/api/v1/namespaces/{namespace}/pods/{name}/eviction",
             data=json.dumps(body),
             headers={"Content-Type": "application/json"},
@@ -248,7 +260,8 @@ class K8sClient:
         }
         if resource_version:
             params["resourceVersion"] = resource_version
-        resp = self._session.get(
+        resp = self._request(
+            "GET",
             f"{self.base_url}/api/v1/nodes",
             params=params,
             stream=True,

@@ -73,6 +73,7 @@ class FakeCluster:
         self.k8s_events: List[Dict[str, Any]] = []
         self._events: List[Dict[str, Any]] = []  # {"rv", "type", "node"}
 
+        self._require_token = ""
         self._server: Optional[ThreadingHTTPServer] = None
         self._server_thread: Optional[threading.Thread] = None
         self._operator_thread: Optional[threading.Thread] = None
@@ -227,8 +228,12 @@ class FakeCluster:
     # ------------------------------------------------------------------
     # HTTP server
     # ------------------------------------------------------------------
-    def start(self) -> str:
+    def start(self, ssl_context=None, require_token: str = "") -> str:
+        """Start the HTTP(S) server. ``ssl_context``: an ``ssl.SSLContext``
+        for TLS (the real in-cluster path); ``require_token``: reject
+        requests without this bearer token (401)."""
         cluster = self
+        self._require_token = require_token
 
         class Handler(BaseHTTPRequestHandler):
             protocol_version = "HTTP/1.1"
@@ -246,7 +251,18 @@ class FakeCluster:
                 self.end_headers()
                 self.wfile.write(body)
 
+            def _auth_ok(self) -> bool:
+                if not cluster._require_token:
+                    return True
+                got = self.headers.get("Authorization", "")
+                if got == f"Bearer {cluster._require_token}":
+                    return True
+                self._send_json(401, {"kind": "Status", "code": 401})
+                return False
+
             def do_GET(self) -> None:
+                if not self._auth_ok():
+                    return
                 url = urlparse(self.path)
                 qs = parse_qs(url.query)
                 parts = [p for p in url.path.split("/") if p]
@@ -275,6 +291,8 @@ class FakeCluster:
                     pass
 
             def do_POST(self) -> None:
+                if not self._auth_ok():
+                    return
                 # pods/eviction subresource: delete the pod after the
                 # configurable delete_delay (graceful termination)
                 url = urlparse(self.path)
@@ -318,6 +336,8 @@ class FakeCluster:
                 self._send_json(404, {"kind": "Status", "code": 404})
 
             def do_PATCH(self) -> None:
+                if not self._auth_ok():
+                    return
                 url = urlparse(self.path)
                 parts = [p for p in url.path.split("/") if p]
                 length = int(self.headers.get("Content-Length", 0))
@@ -437,6 +457,12 @@ class FakeCluster:
 
         self._server = ThreadingHTTPServer(("127.0.0.1", 0), Handler)
         self._server.daemon_threads = True
+        scheme = "http"
+        if ssl_context is not None:
+            self._server.socket = ssl_context.wrap_socket(
+                self._server.socket, server_side=True
+            )
+            scheme = "https"
         self._server_thread = threading.Thread(
             target=self._server.serve_forever, name="fake-apiserver", daemon=True
         )
@@ -446,7 +472,7 @@ class FakeCluster:
         )
         self._operator_thread.start()
         host, port = self._server.server_address
-        return f"http://{host}:{port}"
+        return f"{scheme}://{host}:{port}"
 
     def stop(self) -> None:
         self._stopping = True

@@ -18,7 +18,7 @@ def test_in_cluster_reads_sa_credentials(tmp_path, monkeypatch):
     c = K8sClient.in_cluster()
     assert c.base_url == "https://10.0.0.1:6443"
     assert c._session.headers["Authorization"] == "Bearer sekrit-token"
-    assert c._session.verify == str(sa / "ca.crt")
+    assert c._verify == str(sa / "ca.crt")
 
 
 def test_in_cluster_requires_env(monkeypatch):

@@ -1,0 +1,89 @@
+"""TLS + bearer-token leg of the K8s client — the real in-cluster path
+(HTTPS with a cluster CA and a service-account token)."""
+
+import ssl
+import subprocess
+
+import pytest
+
+from k8s_cc_manager_amd.k8s.client import ApiError, K8sClient
+from k8s_cc_manager_amd.k8s.fakecluster import FakeCluster
+
+
+@pytest.fixture(scope="module")
+def tls_material(tmp_path_factory):
+    d = tmp_path_factory.mktemp("tls")
+    cert, key = d / "cert.pem", d / "key.pem"
+    subprocess.run(
+        [
+            "openssl", "req", "-x509", "-newkey", "rsa:2048", "-nodes",
+            "-keyout", str(key), "-out", str(cert), "-days", "1",
+            "-subj", "/CN=127.0.0.1",
+            "-addext", "subjectAltName=IP:127.0.0.1",
+            # urllib3 v2 requires the trust anchor to be a CA cert
+            "-addext", "basicConstraints=critical,CA:TRUE",
+            "-addext", "keyUsage=keyCertSign,digitalSignature,keyEncipherment",
+        ],
+        check=True,
+        capture_output=True,
+    )
+    return cert, key
+
+
+@pytest.fixture
+def tls_cluster(tls_material):
+    cert, key = tls_material
+    ctx = ssl.SSLContext(ssl.PROTOCOL_TLS_SERVER)
+    ctx.load_cert_chain(str(cert), str(key))
+    cluster = FakeCluster()
+    url = cluster.start(ssl_context=ctx, require_token="sekrit")
+    cluster.add_node("tlsnode", labels={"x": "1"})
+    yield cluster, url, str(cert)
+    cluster.stop()
+
+
+def test_https_with_ca_verify_and_token(tls_cluster):
+    cluster, url, ca = tls_cluster
+    client = K8sClient(url, token="sekrit", verify=ca)
+    node = client.get_node("tlsnode")
+    assert node["metadata"]["labels"]["x"] == "1"
+    client.patch_node_labels("tlsnode", {"y": "2"})
+    assert cluster.node_labels("tlsnode")["y"] == "2"
+
+
+def test_https_rejects_wrong_token(tls_cluster):
+    _, url, ca = tls_cluster
+    client = K8sClient(url, token="wrong", verify=ca)
+    with pytest.raises(ApiError) as ei:
+        client.get_node("tlsnode")
+    assert ei.value.status == 401
+
+
+def test_https_rejects_untrusted_ca(tls_cluster):
+    _, url, _ = tls_cluster
+    import requests
+
+    client = K8sClient(url, token="sekrit", verify=True)  # system CAs only
+    with pytest.raises(requests.exceptions.SSLError):
+        client.get_node("tlsnode")
+
+
+def test_https_watch_stream(tls_cluster):
+    cluster, url, ca = tls_cluster
+    import threading
+    import time
+
+    client = K8sClient(url, token="sekrit", verify=ca)
+    rv = client.get_node("tlsnode")["metadata"]["resourceVersion"]
+    seen = []
+
+    def mutate():
+        time.sleep(0.1)
+        cluster.set_node_label("tlsnode", "watched", "yes")
+
+    threading.Thread(target=mutate).start()
+    for event in client.watch_node("tlsnode", resource_version=rv, timeout_seconds=3):
+        seen.append(event)
+        if event["object"]["metadata"]["labels"].get("watched") == "yes":
+            break
+    assert seen
