@@ -145,6 +145,10 @@ def _load() -> ctypes.CDLL:
     ]
     lib.cc_mfma_gemm_fp8.restype = ctypes.c_int
     lib.cc_mfma_gemm_fp8.argtypes = lib.cc_mfma_gemm_bf16.argtypes
+    lib.cc_mfma_gemm_fp8_variant.restype = ctypes.c_int
+    lib.cc_mfma_gemm_fp8_variant.argtypes = lib.cc_mfma_gemm_bf16.argtypes + [
+        ctypes.c_int
+    ]
     _lib = lib
     return lib
 
@@ -274,6 +278,17 @@ def mfma_gemm_fp8(device_index: int, a_ptr: int, bt_ptr: int, c_ptr: int,
     rc = _load().cc_mfma_gemm_fp8(device_index, a_ptr, bt_ptr, c_ptr, m, n, k)
     if rc != 0:
         raise AttestationError(f"mfma_gemm_fp8 rc={rc}")
+
+
+def mfma_gemm_fp8_variant(device_index: int, a_ptr: int, bt_ptr: int,
+                          c_ptr: int, m: int, n: int, k: int,
+                          which: int) -> None:
+    """Force an fp8 variant: 0=128/BK128, 1=256-deep, 2=128/BK256."""
+    rc = _load().cc_mfma_gemm_fp8_variant(
+        device_index, a_ptr, bt_ptr, c_ptr, m, n, k, which
+    )
+    if rc != 0:
+        raise AttestationError(f"mfma_gemm_fp8_variant({which}) rc={rc}")
 
 
 def mfma_gemm_bf16_variant(device_index: int, a_ptr: int, bt_ptr: int,

@@ -628,6 +628,106 @@ __device__ __forceinline__ void stage_tile_glds8x4(
   }
 }
 
+// fp8 step-3 variant with BK=256: doubles FLOPs per staged byte (the
+// PMC profiles show BOTH fp8 structures park ~53% on data waits with
+// identical MFMA busy — DMA/fetch-rate-bound, so feed each 32 KiB
+// stage twice the math). 128 KiB LDS -> 1 block/CU. The 4-bit swizzle
+// (bank bits 2-5 ^= row bits 3,0,1 + ks) makes ds_read_b128 groups
+// conflict-free over the 256-B-row image; the bit-4 XOR swaps the two
+// 16-B halves of a fragment, which is harmless because A and B use the
+// SAME map (a consistent k-permutation cannot change the dot product).
+constexpr int BK8L = 256;  // fp8 elements per K-tile (256 B rows)
+
+__device__ __forceinline__ int swz256(int off) {
+  return off ^ (((off >> 8) & 3) << 5) ^ (((off >> 10) & 1) << 7) ^
+         (((off >> 11) & 1) << 4);
+}
+
+__device__ __forceinline__ void stage_tile_glds8w(
+    const char* gbase, long row_stride_b, long k0_b, char* lds_tile,
+    int wave, int lane) {
+#pragma unroll
+  for (int p = 0; p < 8; ++p) {
+    int base = (wave * 8 + p) * 1024;
+    int logical = swz256(base + lane * 16);
+    int row = logical >> 8;  // 256 B per row
+    int colb = logical & 255;
+    const char* g = gbase + (long)row * row_stride_b + k0_b + colb;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)g,
+        (__attribute__((address_space(3))) void*)(lds_tile + base), 16, 0, 0);
+  }
+}
+
+__global__ __launch_bounds__(256, 1) void mfma_gemm_fp8_128w(
+    const char* __restrict__ A, const char* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K) {
+  __shared__ char lds[2 * 2 * 32768];  // [buf][A|B][32 KiB]
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wave_m = (wave >> 1) * 64;
+  const int wave_n = (wave & 1) * 64;
+  const int block_m = blockIdx.y * BM;
+  const int block_n = blockIdx.x * BN;
+
+  const char* gA = A + (long)block_m * K;
+  const char* gB = Bt + (long)block_n * K;
+  const long row_b = (long)K;
+
+  f32x4 acc[4][4] = {};
+  const int lane15 = lane & 15;
+  const int kq_b = (lane >> 4) * 32;
+
+  const int nk = K / BK8L;
+  stage_tile_glds8w(gA, row_b, 0, &lds[0], wave, lane);
+  stage_tile_glds8w(gB, row_b, 0, &lds[32768], wave, lane);
+  __syncthreads();
+
+  for (int kt = 0; kt < nk; ++kt) {
+    const int buf = kt & 1;
+    char* As = &lds[buf * 2 * 32768];
+    char* Bs = As + 32768;
+    if (kt + 1 < nk) {
+      char* An = &lds[(buf ^ 1) * 2 * 32768];
+      stage_tile_glds8w(gA, row_b, (long)(kt + 1) * BK8L, An, wave, lane);
+      stage_tile_glds8w(gB, row_b, (long)(kt + 1) * BK8L, An + 32768, wave, lane);
+    }
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      v8i afrag[4], bfrag[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int la = (wave_m + i * 16 + lane15) * 256 + ks * 128 + kq_b;
+        int lb = (wave_n + i * 16 + lane15) * 256 + ks * 128 + kq_b;
+        afrag[i] = load_frag32(As + swz256(la));
+        bfrag[i] = load_frag32(Bs + swz256(lb));
+      }
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+              afrag[i], bfrag[j], acc[i][j], 0, 0, 0, SCALE_ONE, 0, SCALE_ONE);
+    }
+    __syncthreads();
+  }
+
+  const int c_col = lane & 15;
+  const int c_row0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = block_m + wave_m + i * 16 + c_row0 + r;
+        int col = block_n + wave_n + j * 16 + c_col;
+        C[(long)row * N + col] = acc[i][j][r];
+      }
+}
+
 // Deep-pipelined fp8: the same 8-phase 256x256 schedule as the bf16
 // template (identical 16 KiB half-tiles, prefetch map and vmcnt(6)
 // drains — the fp8 image has the same 128-B rows: 128 fp8 elements per
@@ -1160,6 +1260,32 @@ int cc_mfma_gemm_bf16_variant(int device, const void* A, const void* Bt,
     dim3 grid(N / BN, M / BM);
     hipLaunchKernelGGL(mfma_gemm_bf16, grid, dim3(256), 0, 0, (const bf16*)A,
                        (const bf16*)Bt, (float*)C, M, N, K);
+  }
+  return (int)hipDeviceSynchronize();
+}
+
+// Force a specific fp8 variant: 0 = 128-tile BK=128 step-3,
+// 1 = 256-tile deep pipeline, 2 = 128-tile BK=256 step-3.
+int cc_mfma_gemm_fp8_variant(int device, const void* A, const void* Bt,
+                             void* C, int M, int N, int K, int which) {
+  if (hipSetDevice(device) != hipSuccess) return -3;
+  if (which == 2) {
+    if (M % BM || N % BN || K % BK8L) return -2;
+    dim3 grid(N / BN, M / BM);
+    hipLaunchKernelGGL(mfma_gemm_fp8_128w, grid, dim3(256), 0, 0,
+                       (const char*)A, (const char*)Bt, (float*)C, M, N, K);
+  } else if (which == 1) {
+    if (M % BM2 || N % BN2 || K % (2 * BK8)) return -2;
+    dim3 grid(N / BN2, M / BM2);
+    long ws = (long)K * (M + N) + 4L * M * N;
+    hipLaunchKernelGGL(mfma_gemm_fp8_256, grid, dim3(512), 0, 0,
+                       (const char*)A, (const char*)Bt, (float*)C, M, N, K,
+                       ws > (256L << 20) ? 1 : 0);
+  } else {
+    if (M % BM || N % BN || K % BK8) return -2;
+    dim3 grid(N / BN, M / BM);
+    hipLaunchKernelGGL(mfma_gemm_fp8_128, grid, dim3(256), 0, 0,
+                       (const char*)A, (const char*)Bt, (float*)C, M, N, K);
   }
   return (int)hipDeviceSynchronize();
 }
