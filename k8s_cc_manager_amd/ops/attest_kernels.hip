@@ -631,16 +631,17 @@ __device__ __forceinline__ void stage_tile_glds8x4(
 // fp8 step-3 variant with BK=256: doubles FLOPs per staged byte (the
 // PMC profiles show BOTH fp8 structures park ~53% on data waits with
 // identical MFMA busy — DMA/fetch-rate-bound, so feed each 32 KiB
-// stage twice the math). 128 KiB LDS -> 1 block/CU. The 4-bit swizzle
-// (bank bits 2-5 ^= row bits 3,0,1 + ks) makes ds_read_b128 groups
-// conflict-free over the 256-B-row image; the bit-4 XOR swaps the two
-// 16-B halves of a fragment, which is harmless because A and B use the
-// SAME map (a consistent k-permutation cannot change the dot product).
+// stage twice the math). 128 KiB LDS -> 1 block/CU. Swizzle constraint
+// learned the hard way: XOR source bits must address whole fragments
+// (>= 32 B, byte bits >= 5) — a bit-4 XOR keyed on a ROW bit permutes
+// bytes inside the fragment differently for A-row m and B-row n,
+// breaking the MFMA's byte pairing (wrong results, caught by the
+// bitwise check). Bits 5-7 relocate fragments wholesale and are safe;
+// the residual is a 2-way bank conflict on rows r / r+8.
 constexpr int BK8L = 256;  // fp8 elements per K-tile (256 B rows)
 
 __device__ __forceinline__ int swz256(int off) {
-  return off ^ (((off >> 8) & 3) << 5) ^ (((off >> 10) & 1) << 7) ^
-         (((off >> 11) & 1) << 4);
+  return off ^ (((off >> 8) & 3) << 5) ^ (((off >> 10) & 1) << 7);
 }
 
 __device__ __forceinline__ void stage_tile_glds8w(
