@@ -628,6 +628,111 @@ __device__ __forceinline__ void stage_tile_glds8x4(
   }
 }
 
+// fp8 step-3 variant with BK=64 + the 32x32x64 op: 32 KiB LDS/block
+// -> 4 blocks/CU (16 waves/CU) — tests the occupancy hypothesis (the
+// BK=256 arm showed the limiter is latency hiding, not FLOP/byte).
+// Image rows are 64 B; swizzle injects row bit 2 into bank bit 3
+// (byte 5 — whole-fragment granularity, see the BK=256 lesson);
+// residual 4-way conflicts on rows mod 8 accepted for the experiment.
+__device__ __forceinline__ int swz64(int off) {
+  return off ^ (((off >> 8) & 1) << 5);
+}
+
+__device__ __forceinline__ void stage_tile_glds8s(
+    const char* gbase, long row_stride_b, long k0_b, char* lds_tile,
+    int wave, int lane) {
+#pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    int base = (wave * 2 + p) * 1024;
+    int logical = swz64(base + lane * 16);
+    int row = logical >> 6;  // 64 B per row
+    int colb = logical & 63;
+    const char* g = gbase + (long)row * row_stride_b + k0_b + colb;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)g,
+        (__attribute__((address_space(3))) void*)(lds_tile + base), 16, 0, 0);
+  }
+}
+
+#define MFMA_FP8W_ASM(ACCX, AOP, BOP)                                         \
+  asm volatile(                                                               \
+      "v_mfma_scale_f32_32x32x64_f8f6f4 %0, %1, %2, %0, %3, %4 "              \
+      "op_sel_hi:[0,0,0]"                                                     \
+      : "+v"(ACCX)                                                            \
+      : "v"(AOP), "v"(BOP), "v"(sc_reg), "v"(sc_reg))
+
+__global__ __launch_bounds__(256, 4) void mfma_gemm_fp8_128s(
+    const char* __restrict__ A, const char* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K) {
+  __shared__ char lds[2 * 2 * 8192];  // [buf][A|B][8 KiB]
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wave_m = (wave >> 1) * 64;
+  const int wave_n = (wave & 1) * 64;
+  const int block_m = blockIdx.y * BM;
+  const int block_n = blockIdx.x * BN;
+
+  const char* gA = A + (long)block_m * K;
+  const char* gB = Bt + (long)block_n * K;
+  const long row_b = (long)K;
+
+  typedef __attribute__((ext_vector_type(16))) float f32x16v;
+  f32x16v acc[2][2] = {};  // 2x2 tiles of 32x32 per wave (64 VGPR)
+  const int lane31 = lane & 31;
+  const int kq_b = (lane >> 5) * 32;
+  int sc_reg;
+  asm("v_mov_b32 %0, 0x7f7f7f7f" : "=v"(sc_reg));
+
+  const int nk = K / 64;
+  stage_tile_glds8s(gA, row_b, 0, &lds[0], wave, lane);
+  stage_tile_glds8s(gB, row_b, 0, &lds[8192], wave, lane);
+  __syncthreads();
+
+  for (int kt = 0; kt < nk; ++kt) {
+    const int buf = kt & 1;
+    char* As = &lds[buf * 2 * 8192];
+    char* Bs = As + 8192;
+    if (kt + 1 < nk) {
+      char* An = &lds[(buf ^ 1) * 2 * 8192];
+      stage_tile_glds8s(gA, row_b, (long)(kt + 1) * 64, An, wave, lane);
+      stage_tile_glds8s(gB, row_b, (long)(kt + 1) * 64, An + 8192, wave, lane);
+    }
+    {
+      v8i afrag[2], bfrag[2];
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        int la = (wave_m + i * 32 + lane31) * 64 + kq_b;
+        int lb = (wave_n + i * 32 + lane31) * 64 + kq_b;
+        afrag[i] = load_frag32(As + swz64(la));
+        bfrag[i] = load_frag32(Bs + swz64(lb));
+      }
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+          MFMA_FP8W_ASM(acc[i][j], afrag[i], bfrag[j]);
+    }
+    __syncthreads();
+  }
+  asm volatile("s_nop 15\ns_nop 15\ns_nop 2" :::);
+
+  // 32x32 C/D map: col=lane&31, row=(reg&3)+8*(reg>>2)+4*(lane>>5)
+  const int c_col32 = lane & 31;
+  const int c_rowhi = (lane >> 5) * 4;
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+#pragma unroll
+      for (int reg = 0; reg < 16; ++reg) {
+        int row = block_m + wave_m + i * 32 + (reg & 3) + 8 * (reg >> 2) + c_rowhi;
+        int col = block_n + wave_n + j * 32 + c_col32;
+        C[(long)row * N + col] = acc[i][j][reg];
+      }
+}
+
 // fp8 step-3 variant with BK=256: doubles FLOPs per staged byte (the
 // PMC profiles show BOTH fp8 structures park ~53% on data waits with
 // identical MFMA busy — DMA/fetch-rate-bound, so feed each 32 KiB
@@ -1270,7 +1375,12 @@ int cc_mfma_gemm_bf16_variant(int device, const void* A, const void* Bt,
 int cc_mfma_gemm_fp8_variant(int device, const void* A, const void* Bt,
                              void* C, int M, int N, int K, int which) {
   if (hipSetDevice(device) != hipSuccess) return -3;
-  if (which == 2) {
+  if (which == 3) {
+    if (M % BM || N % BN || K % 64) return -2;
+    dim3 grid(N / BN, M / BM);
+    hipLaunchKernelGGL(mfma_gemm_fp8_128s, grid, dim3(256), 0, 0,
+                       (const char*)A, (const char*)Bt, (float*)C, M, N, K);
+  } else if (which == 2) {
     if (M % BM || N % BN || K % BK8L) return -2;
     dim3 grid(N / BN, M / BM);
     hipLaunchKernelGGL(mfma_gemm_fp8_128w, grid, dim3(256), 0, 0,
