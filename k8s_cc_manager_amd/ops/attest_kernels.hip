@@ -1406,12 +1406,13 @@ int cc_mfma_gemm_fp8_variant(int device, const void* A, const void* Bt,
 int cc_mfma_gemm_fp8(int device, const void* A, const void* Bt, void* C,
                      int M, int N, int K) {
   if (hipSetDevice(device) != hipSuccess) return -3;
-  if (M % BM2 == 0 && N % BN2 == 0 && K % (2 * BK8) == 0) {
-    dim3 grid(N / BN2, M / BM2);
-    long ws = (long)K * (M + N) + 4L * M * N;
-    hipLaunchKernelGGL(mfma_gemm_fp8_256, grid, dim3(512), 0, 0,
-                       (const char*)A, (const char*)Bt, (float*)C, M, N, K,
-                       ws > (256L << 20) ? 1 : 0);
+  // measured A/B (profiles/, 2026-09-13): the 4-blocks/CU BK=64 shape
+  // wins past ~32M output elements (1586 vs 1505 TF @8192^3); the
+  // 2-blocks/CU BK=128 shape wins below (1375 vs 1353 @4096^3).
+  if ((long)M * N > (32L << 20) && M % BM == 0 && N % BN == 0 && K % 64 == 0) {
+    dim3 grid(N / BN, M / BM);
+    hipLaunchKernelGGL(mfma_gemm_fp8_128s, grid, dim3(256), 0, 0,
+                       (const char*)A, (const char*)Bt, (float*)C, M, N, K);
     return (int)hipDeviceSynchronize();
   }
   if (M % BM || N % BN || K % BK8) return -2;
