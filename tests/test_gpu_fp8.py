@@ -53,6 +53,25 @@ def test_fp8_deep_kernel_race_screen(attest):
         assert torch.equal(c, ref), f"trial {trial}"
 
 
+@pytest.mark.parametrize("which", [0, 2, 3])
+def test_fp8_variants_bitwise(attest, which):
+    """Every fp8 structure variant (BK=128 / BK=256 / BK=64-hiocc) must
+    agree bitwise with the fp32 reference on integer data."""
+    m = n = 512
+    k = 1024
+    torch.manual_seed(which)
+    a = _fp8(torch.randint(-2, 2, (m, k), device="cuda").float())
+    bt = _fp8(torch.randint(-2, 2, (n, k), device="cuda").float())
+    ref = a.float() @ bt.float().t()
+    for trial in range(5):
+        c = torch.empty(m, n, device="cuda", dtype=torch.float32)
+        attest.mfma_gemm_fp8_variant(
+            0, a.data_ptr(), bt.data_ptr(), c.data_ptr(), m, n, k, which
+        )
+        torch.cuda.synchronize()
+        assert torch.equal(c, ref), f"which={which} trial={trial}"
+
+
 def test_fp8_step3_kernel_still_used_for_small_shapes(attest):
     """K%256 != 0 routes to the 128-tile step-3 kernel."""
     m, n, k = 128, 128, 384
