@@ -175,8 +175,28 @@ class AmdSmiDevice(CCDevice):
         try:
             reset_node.write_text("1")
             logger.info("%s: sysfs FLR issued", self.bdf)
+            return
         except OSError as e:
-            raise ResetError(f"{self.bdf}: reset failed: {e}")
+            logger.warning("%s: sysfs FLR failed (%s)", self.bdf, e)
+        # last-resort escalation: full amdgpu driver reload. NODE-WIDE
+        # (affects every GPU), so doubly gated: CC_MANAGER_ALLOW_RESET
+        # (we are inside it) AND CC_ALLOW_DRIVER_RELOAD=1.
+        if os.environ.get("CC_ALLOW_DRIVER_RELOAD", "0") == "1":
+            try:
+                import amdsmi
+
+                amdsmi.amdsmi_gpu_driver_reload()
+                logger.warning("%s: escalated to amdgpu driver reload", self.bdf)
+                return
+            except Exception as e:
+                raise ResetError(
+                    f"{self.bdf}: reset failed at every tier "
+                    f"(amdsmi, FLR, driver reload: {e})"
+                )
+        raise ResetError(
+            f"{self.bdf}: reset failed (amdsmi and sysfs FLR; driver "
+            "reload not permitted — set CC_ALLOW_DRIVER_RELOAD=1)"
+        )
 
     def wait_for_boot(self, timeout: float = 60.0) -> None:
         """Poll until the device answers amdsmi queries AND (when the

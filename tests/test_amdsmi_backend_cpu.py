@@ -90,3 +90,54 @@ def test_get_backend_auto_falls_back_to_mock(monkeypatch, tmp_path):
     monkeypatch.setenv("CC_STATE_DIR", str(tmp_path))
     be = get_backend("auto", num_gpus=2)
     assert isinstance(be, MockBackend) or be.find_devices()[1] > 0
+
+
+def _reset_device(tmp_path, monkeypatch, fake_amdsmi):
+    """AmdSmiDevice with allow_reset and a stubbed amdsmi module."""
+    import sys
+
+    monkeypatch.setitem(sys.modules, "amdsmi", fake_amdsmi)
+    dev, store = _device(tmp_path, allow_reset=True)
+    return dev
+
+
+def test_hard_reset_amdsmi_first(tmp_path, monkeypatch):
+    import types
+
+    calls = []
+    fake = types.SimpleNamespace(
+        amdsmi_reset_gpu=lambda h: calls.append("amdsmi_reset"),
+    )
+    dev = _reset_device(tmp_path, monkeypatch, fake)
+    dev.reset()
+    assert calls == ["amdsmi_reset"]
+
+
+def test_hard_reset_escalates_to_driver_reload(tmp_path, monkeypatch):
+    """amdsmi reset fails, sysfs FLR fails (no such bdf on this box) ->
+    driver reload only with CC_ALLOW_DRIVER_RELOAD=1."""
+    import types
+
+    import pytest as _pytest
+
+    from k8s_cc_manager_amd.device.contract import ResetError
+
+    calls = []
+
+    def failing_reset(h):
+        raise RuntimeError("injected amdsmi failure")
+
+    fake = types.SimpleNamespace(
+        amdsmi_reset_gpu=failing_reset,
+        amdsmi_gpu_driver_reload=lambda: calls.append("driver_reload"),
+    )
+    monkeypatch.delenv("CC_ALLOW_DRIVER_RELOAD", raising=False)
+    dev = _reset_device(tmp_path, monkeypatch, fake)
+    with _pytest.raises(ResetError, match="driver"):
+        dev.reset()
+    assert calls == []
+
+    monkeypatch.setenv("CC_ALLOW_DRIVER_RELOAD", "1")
+    dev2 = _reset_device(tmp_path, monkeypatch, fake)
+    dev2.reset()
+    assert calls == ["driver_reload"]
