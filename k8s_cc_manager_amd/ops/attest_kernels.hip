@@ -733,6 +733,80 @@ __global__ __launch_bounds__(256, 4) void mfma_gemm_fp8_128s(
       }
 }
 
+// fp8 BK=128 at 3 blocks/CU via 1.5-buffering: B double-buffered
+// (prefetch overlaps compute), A single-buffered (restaged in a serial
+// window between two barriers) -> 48 KiB LDS/block. Tests the middle
+// point of the occupancy curve (2 blocks: 1505 TF, 4 blocks: 1586).
+__global__ __launch_bounds__(256, 3) void mfma_gemm_fp8_128t(
+    const char* __restrict__ A, const char* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K) {
+  __shared__ char lds[3 * 16384];  // [A][B0][B1]
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wave_m = (wave >> 1) * 64;
+  const int wave_n = (wave & 1) * 64;
+  const int block_m = blockIdx.y * BM;
+  const int block_n = blockIdx.x * BN;
+
+  const char* gA = A + (long)block_m * K;
+  const char* gB = Bt + (long)block_n * K;
+  const long row_b = (long)K;
+
+  f32x4 acc[4][4] = {};
+  const int lane15 = lane & 15;
+  const int kq_b = (lane >> 4) * 32;
+
+  const int nk = K / BK8;
+  char* As = &lds[0];
+  stage_tile_glds8x4(gA, row_b, 0, As, wave, lane);
+  stage_tile_glds8x4(gB, row_b, 0, &lds[16384], wave, lane);
+  __syncthreads();
+
+  for (int kt = 0; kt < nk; ++kt) {
+    char* Bs = &lds[16384 * (1 + (kt & 1))];
+    if (kt + 1 < nk)  // B prefetch overlaps this tile's compute
+      stage_tile_glds8x4(gB, row_b, (long)(kt + 1) * BK8,
+                         &lds[16384 * (1 + ((kt + 1) & 1))], wave, lane);
+    {
+      v8i afrag[4], bfrag[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int la = (wave_m + i * 16 + lane15) * 128 + kq_b;
+        int lb = (wave_n + i * 16 + lane15) * 128 + kq_b;
+        afrag[i] = load_frag32(As + swz8(la));
+        bfrag[i] = load_frag32(Bs + swz8(lb));
+      }
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+              afrag[i], bfrag[j], acc[i][j], 0, 0, 0, SCALE_ONE, 0, SCALE_ONE);
+    }
+    __syncthreads();  // all reads of As done (drains B prefetch too)
+    if (kt + 1 < nk) {
+      // serial A window: restage in place, land before next compute
+      stage_tile_glds8x4(gA, row_b, (long)(kt + 1) * BK8, As, wave, lane);
+      __syncthreads();
+    }
+  }
+
+  const int c_col = lane & 15;
+  const int c_row0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = block_m + wave_m + i * 16 + c_row0 + r;
+        int col = block_n + wave_n + j * 16 + c_col;
+        C[(long)row * N + col] = acc[i][j][r];
+      }
+}
+
 // fp8 step-3 variant with BK=256: doubles FLOPs per staged byte (the
 // PMC profiles show BOTH fp8 structures park ~53% on data waits with
 // identical MFMA busy — DMA/fetch-rate-bound, so feed each 32 KiB
@@ -1375,7 +1449,12 @@ int cc_mfma_gemm_bf16_variant(int device, const void* A, const void* Bt,
 int cc_mfma_gemm_fp8_variant(int device, const void* A, const void* Bt,
                              void* C, int M, int N, int K, int which) {
   if (hipSetDevice(device) != hipSuccess) return -3;
-  if (which == 3) {
+  if (which == 4) {
+    if (M % BM || N % BN || K % BK8) return -2;
+    dim3 grid(N / BN, M / BM);
+    hipLaunchKernelGGL(mfma_gemm_fp8_128t, grid, dim3(256), 0, 0,
+                       (const char*)A, (const char*)Bt, (float*)C, M, N, K);
+  } else if (which == 3) {
     if (M % BM || N % BN || K % 64) return -2;
     dim3 grid(N / BN, M / BM);
     hipLaunchKernelGGL(mfma_gemm_fp8_128s, grid, dim3(256), 0, 0,
