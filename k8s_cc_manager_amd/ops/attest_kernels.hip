@@ -1485,14 +1485,22 @@ int cc_mfma_gemm_fp8_variant(int device, const void* A, const void* Bt,
 int cc_mfma_gemm_fp8(int device, const void* A, const void* Bt, void* C,
                      int M, int N, int K) {
   if (hipSetDevice(device) != hipSuccess) return -3;
-  // measured A/B (profiles/, 2026-09-13): the 4-blocks/CU BK=64 shape
-  // wins past ~32M output elements (1586 vs 1505 TF @8192^3); the
-  // 2-blocks/CU BK=128 shape wins below (1375 vs 1353 @4096^3).
-  if ((long)M * N > (32L << 20) && M % BM == 0 && N % BN == 0 && K % 64 == 0) {
+  // measured A/B (profiles/, 2026-09-13): past ~32M output elements
+  // occupancy dominates — the 3-blocks/CU 1.5-buffered BK=128 shape
+  // wins (1682 vs 1597 bk64-4blk vs 1530 bk128-2blk TF @8192^3); the
+  // 2-blocks/CU BK=128 shape wins below (1395 vs 1318/1085 @4096^3).
+  if ((long)M * N > (32L << 20) && M % BM == 0 && N % BN == 0) {
     dim3 grid(N / BN, M / BM);
-    hipLaunchKernelGGL(mfma_gemm_fp8_128s, grid, dim3(256), 0, 0,
-                       (const char*)A, (const char*)Bt, (float*)C, M, N, K);
-    return (int)hipDeviceSynchronize();
+    if (K % BK8 == 0) {
+      hipLaunchKernelGGL(mfma_gemm_fp8_128t, grid, dim3(256), 0, 0,
+                         (const char*)A, (const char*)Bt, (float*)C, M, N, K);
+      return (int)hipDeviceSynchronize();
+    }
+    if (K % 64 == 0) {
+      hipLaunchKernelGGL(mfma_gemm_fp8_128s, grid, dim3(256), 0, 0,
+                         (const char*)A, (const char*)Bt, (float*)C, M, N, K);
+      return (int)hipDeviceSynchronize();
+    }
   }
   if (M % BM || N % BN || K % BK8) return -2;
   dim3 grid(N / BN, M / BM);
