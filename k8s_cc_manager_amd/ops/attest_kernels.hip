@@ -733,6 +733,78 @@ __global__ __launch_bounds__(256, 4) void mfma_gemm_fp8_128s(
       }
 }
 
+// fp8 BK=128 single-buffered at 4 blocks/CU: the remaining untried
+// corner of the (BK, blocks/CU) grid — same 32 KiB LDS as the BK=64
+// double-buffered shape (128s) but spent on tile DEPTH instead of a
+// second buffer: stage A+B serially, one sync, 2x the math per
+// barrier pair; the serial stage window is covered by the other 3
+// blocks on the CU. Uses the 32x32x64 asm op (acc 2x2xf32x16 = 64
+// VGPR) to stay under the 128-VGPR budget of 4 waves/SIMD.
+__global__ __launch_bounds__(256, 4) void mfma_gemm_fp8_128u(
+    const char* __restrict__ A, const char* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K) {
+  __shared__ char lds[2 * 16384];  // [A][B], single-buffered
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wave_m = (wave >> 1) * 64;
+  const int wave_n = (wave & 1) * 64;
+  const int block_m = blockIdx.y * BM;
+  const int block_n = blockIdx.x * BN;
+
+  const char* gA = A + (long)block_m * K;
+  const char* gB = Bt + (long)block_n * K;
+  const long row_b = (long)K;
+
+  typedef __attribute__((ext_vector_type(16))) float f32x16v;
+  f32x16v acc[2][2] = {};
+  const int lane31 = lane & 31;
+  const int kq_b = (lane >> 5) * 32;
+  int sc_reg;
+  asm("v_mov_b32 %0, 0x7f7f7f7f" : "=v"(sc_reg));
+
+  char* As = &lds[0];
+  char* Bs = &lds[16384];
+  const int nk = K / BK8;
+  for (int kt = 0; kt < nk; ++kt) {
+    stage_tile_glds8x4(gA, row_b, (long)kt * BK8, As, wave, lane);
+    stage_tile_glds8x4(gB, row_b, (long)kt * BK8, Bs, wave, lane);
+    __syncthreads();
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      v8i afrag[2], bfrag[2];
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        int la = (wave_m + i * 32 + lane31) * 128 + ks * 64 + kq_b;
+        int lb = (wave_n + i * 32 + lane31) * 128 + ks * 64 + kq_b;
+        afrag[i] = load_frag32(As + swz8(la));
+        bfrag[i] = load_frag32(Bs + swz8(lb));
+      }
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+          MFMA_FP8W_ASM(acc[i][j], afrag[i], bfrag[j]);
+    }
+    __syncthreads();
+  }
+  asm volatile("s_nop 15\ns_nop 15\ns_nop 2" :::);
+
+  const int c_col32 = lane & 31;
+  const int c_rowhi = (lane >> 5) * 4;
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+#pragma unroll
+      for (int reg = 0; reg < 16; ++reg) {
+        int row = block_m + wave_m + i * 32 + (reg & 3) + 8 * (reg >> 2) + c_rowhi;
+        int col = block_n + wave_n + j * 32 + c_col32;
+        C[(long)row * N + col] = acc[i][j][reg];
+      }
+}
+
 // fp8 BK=128 at 3 blocks/CU via 1.5-buffering: B double-buffered
 // (prefetch overlaps compute), A single-buffered (restaged in a serial
 // window between two barriers) -> 48 KiB LDS/block. Tests the middle
@@ -1449,7 +1521,12 @@ int cc_mfma_gemm_bf16_variant(int device, const void* A, const void* Bt,
 int cc_mfma_gemm_fp8_variant(int device, const void* A, const void* Bt,
                              void* C, int M, int N, int K, int which) {
   if (hipSetDevice(device) != hipSuccess) return -3;
-  if (which == 4) {
+  if (which == 5) {
+    if (M % BM || N % BN || K % BK8) return -2;
+    dim3 grid(N / BN, M / BM);
+    hipLaunchKernelGGL(mfma_gemm_fp8_128u, grid, dim3(256), 0, 0,
+                       (const char*)A, (const char*)Bt, (float*)C, M, N, K);
+  } else if (which == 4) {
     if (M % BM || N % BN || K % BK8) return -2;
     dim3 grid(N / BN, M / BM);
     hipLaunchKernelGGL(mfma_gemm_fp8_128t, grid, dim3(256), 0, 0,
