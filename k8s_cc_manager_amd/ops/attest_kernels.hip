@@ -805,6 +805,74 @@ __global__ __launch_bounds__(256, 4) void mfma_gemm_fp8_128u(
       }
 }
 
+// bf16 transplant of the fp8 winner above: 128x128 tile, BK=64
+// (128-B rows, so the same 16-KiB-tile stager and swz8 image apply
+// byte-for-byte), single-buffered 32 KiB LDS -> 4 blocks/CU, 32x32x16
+// ops with 2x2 tiles/wave (acc 64 VGPR). Tests whether the occupancy
+// lever that took fp8 from 1516 to 2052 TF also moves bf16.
+__global__ __launch_bounds__(256, 4) void mfma_gemm_bf16_128u(
+    const bf16* __restrict__ A, const bf16* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K) {
+  __shared__ char lds[2 * 16384];  // [A][B], single-buffered
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wave_m = (wave >> 1) * 64;
+  const int wave_n = (wave & 1) * 64;
+  const int block_m = blockIdx.y * BM;
+  const int block_n = blockIdx.x * BN;
+
+  const char* gA = (const char*)(A + (long)block_m * K);
+  const char* gB = (const char*)(Bt + (long)block_n * K);
+  const long row_b = (long)K * 2;
+
+  typedef __attribute__((ext_vector_type(16))) float f32x16v;
+  f32x16v acc[2][2] = {};
+  const int lane31 = lane & 31;
+  const int khalf_b = (lane >> 5) * 16;  // (l>>5)*8 bf16 = 16 B
+
+  char* As = &lds[0];
+  char* Bs = &lds[16384];
+  const int nk = K / 64;  // 64 bf16 = 128 B per K-tile
+  for (int kt = 0; kt < nk; ++kt) {
+    stage_tile_glds8x4(gA, row_b, (long)kt * 128, As, wave, lane);
+    stage_tile_glds8x4(gB, row_b, (long)kt * 128, Bs, wave, lane);
+    __syncthreads();
+#pragma unroll
+    for (int ks = 0; ks < 4; ++ks) {
+      bf16x8 af[2], bf[2];
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        int la = (wave_m + i * 32 + lane31) * 128 + ks * 32 + khalf_b;
+        int lb = (wave_n + i * 32 + lane31) * 128 + ks * 32 + khalf_b;
+        af[i] = *(const bf16x8*)(As + swz8(la));
+        bf[i] = *(const bf16x8*)(Bs + swz8(lb));
+      }
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              af[i], bf[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  const int c_col32 = lane & 31;
+  const int c_rowhi = (lane >> 5) * 4;
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+#pragma unroll
+      for (int reg = 0; reg < 16; ++reg) {
+        int row = block_m + wave_m + i * 32 + (reg & 3) + 8 * (reg >> 2) + c_rowhi;
+        int col = block_n + wave_n + j * 32 + c_col32;
+        C[(long)row * N + col] = acc[i][j][reg];
+      }
+}
+
 // fp8 BK=128 at 3 blocks/CU via 1.5-buffering: B double-buffered
 // (prefetch overlaps compute), A single-buffered (restaged in a serial
 // window between two barriers) -> 48 KiB LDS/block. Tests the middle
@@ -1494,7 +1562,12 @@ int cc_mfma_gemm_bf16(int device, const void* A, const void* Bt, void* C,
 int cc_mfma_gemm_bf16_variant(int device, const void* A, const void* Bt,
                               void* C, int M, int N, int K, int which) {
   if (hipSetDevice(device) != hipSuccess) return -3;
-  if (which == 1 || which == 2) {
+  if (which == 3) {
+    if (M % BM || N % BN || K % 64) return -2;
+    dim3 grid(N / BN, M / BM);
+    hipLaunchKernelGGL(mfma_gemm_bf16_128u, grid, dim3(256), 0, 0,
+                       (const bf16*)A, (const bf16*)Bt, (float*)C, M, N, K);
+  } else if (which == 1 || which == 2) {
     if (M % BM2 || N % BN2 || K % (2 * BK2)) return -2;
     dim3 grid(N / BN2, M / BM2);
     long ws = 2L * K * (M + N) + 4L * M * N;
@@ -1562,28 +1635,25 @@ int cc_mfma_gemm_fp8_variant(int device, const void* A, const void* Bt,
 int cc_mfma_gemm_fp8(int device, const void* A, const void* Bt, void* C,
                      int M, int N, int K) {
   if (hipSetDevice(device) != hipSuccess) return -3;
-  // measured A/B (profiles/, 2026-09-13): past ~32M output elements
-  // occupancy dominates — the 3-blocks/CU 1.5-buffered BK=128 shape
-  // wins (1682 vs 1597 bk64-4blk vs 1530 bk128-2blk TF @8192^3); the
-  // 2-blocks/CU BK=128 shape wins below (1395 vs 1318/1085 @4096^3).
-  if ((long)M * N > (32L << 20) && M % BM == 0 && N % BN == 0) {
-    dim3 grid(N / BN, M / BM);
-    if (K % BK8 == 0) {
-      hipLaunchKernelGGL(mfma_gemm_fp8_128t, grid, dim3(256), 0, 0,
-                         (const char*)A, (const char*)Bt, (float*)C, M, N, K);
-      return (int)hipDeviceSynchronize();
-    }
-    if (K % 64 == 0) {
-      hipLaunchKernelGGL(mfma_gemm_fp8_128s, grid, dim3(256), 0, 0,
-                         (const char*)A, (const char*)Bt, (float*)C, M, N, K);
-      return (int)hipDeviceSynchronize();
-    }
-  }
-  if (M % BM || N % BN || K % BK8) return -2;
+  // measured sweep (profiles/fp8_dispatch_sweep.log, 2026-09-13): the
+  // single-buffered 4-blocks/CU BK=128 shape wins or ties at EVERY
+  // size (512..8192^3; 2052 vs 1516 TF @8192^3, 1657 vs 1462 @4096^3
+  // against the 2-blocks/CU double-buffered shape) — occupancy
+  // dominates intra-block overlap on this part. K%64-only shapes fall
+  // back to the BK=64 4-blocks/CU kernel.
+  if (M % BM || N % BN) return -2;
   dim3 grid(N / BN, M / BM);
-  hipLaunchKernelGGL(mfma_gemm_fp8_128, grid, dim3(256), 0, 0, (const char*)A,
-                     (const char*)Bt, (float*)C, M, N, K);
-  return (int)hipDeviceSynchronize();
+  if (K % BK8 == 0) {
+    hipLaunchKernelGGL(mfma_gemm_fp8_128u, grid, dim3(256), 0, 0,
+                       (const char*)A, (const char*)Bt, (float*)C, M, N, K);
+    return (int)hipDeviceSynchronize();
+  }
+  if (K % 64 == 0) {
+    hipLaunchKernelGGL(mfma_gemm_fp8_128s, grid, dim3(256), 0, 0,
+                       (const char*)A, (const char*)Bt, (float*)C, M, N, K);
+    return (int)hipDeviceSynchronize();
+  }
+  return -2;
 }
 
 // Plain fp32 reference GEMM on the same operand convention.

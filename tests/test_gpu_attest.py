@@ -82,6 +82,23 @@ def test_mfma_gemm_256_template_matches_torch(attest, m, n, k, which):
         assert err <= 2e-3 * max(scale, 1.0), f"trial {trial} which={which}: err={err}"
 
 
+def test_mfma_gemm_bf16_4blk_bitwise(attest):
+    """The 4-blocks/CU single-buffered bf16 variant (which=3): bitwise
+    integer race screen (serial stage window discipline)."""
+    m = n = 512
+    k = 1024
+    a = (torch.randint(-2, 2, (m, k), device="cuda")).bfloat16()
+    bt = (torch.randint(-2, 2, (n, k), device="cuda")).bfloat16()
+    ref = a.float() @ bt.float().t()
+    for trial in range(5):
+        c = torch.empty(m, n, device="cuda", dtype=torch.float32)
+        attest.mfma_gemm_bf16_variant(
+            0, a.data_ptr(), bt.data_ptr(), c.data_ptr(), m, n, k, 3
+        )
+        torch.cuda.synchronize()
+        assert torch.equal(c, ref), f"trial {trial} mismatch"
+
+
 def test_mfma_gemm_256_integer_exact_race_screen(attest):
     """Integer data: any stale-LDS race shows as a bitwise mismatch."""
     m = n = 512
