@@ -283,7 +283,9 @@ def mfma_gemm_fp8(device_index: int, a_ptr: int, bt_ptr: int, c_ptr: int,
 def mfma_gemm_fp8_variant(device_index: int, a_ptr: int, bt_ptr: int,
                           c_ptr: int, m: int, n: int, k: int,
                           which: int) -> None:
-    """Force an fp8 variant: 0=128/BK128, 1=256-deep, 2=128/BK256."""
+    """Force an fp8 variant: 0=128/BK128 2-blk, 1=256-deep, 2=128/BK256,
+    3=BK64 4-blk, 4=BK128 3-blk 1.5-buf, 5=BK128 4-blk 1-buf (default
+    dispatch), 6=256x128 tile 2-blk."""
     rc = _load().cc_mfma_gemm_fp8_variant(
         device_index, a_ptr, bt_ptr, c_ptr, m, n, k, which
     )
@@ -294,7 +296,8 @@ def mfma_gemm_fp8_variant(device_index: int, a_ptr: int, bt_ptr: int,
 def mfma_gemm_bf16_variant(device_index: int, a_ptr: int, bt_ptr: int,
                            c_ptr: int, m: int, n: int, k: int,
                            which: int) -> None:
-    """Force a GEMM variant: 0 = 128x128 step-3, 1 = 256x256 8-phase."""
+    """Force a GEMM variant: 0 = 128x128 step-3, 1 = 256x256 8-phase,
+    2 = 256x256 8-phase 32x32-op, 3 = 128x128 4-blocks/CU 1-buf."""
     rc = _load().cc_mfma_gemm_bf16_variant(
         device_index, a_ptr, bt_ptr, c_ptr, m, n, k, which
     )

@@ -805,6 +805,78 @@ __global__ __launch_bounds__(256, 4) void mfma_gemm_fp8_128u(
       }
 }
 
+// fp8 256x128 tile at the winner's wave occupancy: 512 threads,
+// single-buffered 48 KiB LDS (A 32 KiB + B 16 KiB) -> 2 blocks/CU =
+// 16 waves/CU, same as the 128x128 winner, but 1.33x the FLOPs per
+// staged byte (A bytes amortized over twice the output rows). Tests
+// whether the stage window is DMA-rate-bound (this wins) or
+// latency-bound (occupancy already covers it; this ties or loses to
+// the coarser 8-wave barrier).
+__global__ __launch_bounds__(512, 2) void mfma_gemm_fp8_256u(
+    const char* __restrict__ A, const char* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K) {
+  __shared__ char lds[3 * 16384];  // [A: 32 KiB][B: 16 KiB]
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;  // 0..7
+  const int wave_m = (wave >> 1) * 64;  // 0..192
+  const int wave_n = (wave & 1) * 64;   // 0,64
+  const int block_m = blockIdx.y * 256;
+  const int block_n = blockIdx.x * BN;
+
+  const char* gA = A + (long)block_m * K;
+  const char* gB = Bt + (long)block_n * K;
+  const long row_b = (long)K;
+
+  typedef __attribute__((ext_vector_type(16))) float f32x16v;
+  f32x16v acc[2][2] = {};
+  const int lane31 = lane & 31;
+  const int kq_b = (lane >> 5) * 32;
+  int sc_reg;
+  asm("v_mov_b32 %0, 0x7f7f7f7f" : "=v"(sc_reg));
+
+  char* As = &lds[0];
+  char* Bs = &lds[32768];
+  const int nk = K / BK8;
+  for (int kt = 0; kt < nk; ++kt) {
+    stage_tile_glds8x4(gA, row_b, (long)kt * BK8, As, wave, lane);
+    stage_half_glds8(gB, row_b, (long)kt * BK8, 0, Bs, wave, lane);
+    __syncthreads();
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      v8i afrag[2], bfrag[2];
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        int la = (wave_m + i * 32 + lane31) * 128 + ks * 64 + kq_b;
+        int lb = (wave_n + i * 32 + lane31) * 128 + ks * 64 + kq_b;
+        afrag[i] = load_frag32(As + swz8(la));
+        bfrag[i] = load_frag32(Bs + swz8(lb));
+      }
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+          MFMA_FP8W_ASM(acc[i][j], afrag[i], bfrag[j]);
+    }
+    __syncthreads();
+  }
+  asm volatile("s_nop 15\ns_nop 15\ns_nop 2" :::);
+
+  const int c_col32 = lane & 31;
+  const int c_rowhi = (lane >> 5) * 4;
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+#pragma unroll
+      for (int reg = 0; reg < 16; ++reg) {
+        int row = block_m + wave_m + i * 32 + (reg & 3) + 8 * (reg >> 2) + c_rowhi;
+        int col = block_n + wave_n + j * 32 + c_col32;
+        C[(long)row * N + col] = acc[i][j][reg];
+      }
+}
+
 // bf16 transplant of the fp8 winner above: 128x128 tile, BK=64
 // (128-B rows, so the same 16-KiB-tile stager and swz8 image apply
 // byte-for-byte), single-buffered 32 KiB LDS -> 4 blocks/CU, 32x32x16
@@ -1594,7 +1666,12 @@ int cc_mfma_gemm_bf16_variant(int device, const void* A, const void* Bt,
 int cc_mfma_gemm_fp8_variant(int device, const void* A, const void* Bt,
                              void* C, int M, int N, int K, int which) {
   if (hipSetDevice(device) != hipSuccess) return -3;
-  if (which == 5) {
+  if (which == 6) {
+    if (M % 256 || N % BN || K % BK8) return -2;
+    dim3 grid(N / BN, M / 256);
+    hipLaunchKernelGGL(mfma_gemm_fp8_256u, grid, dim3(512), 0, 0,
+                       (const char*)A, (const char*)Bt, (float*)C, M, N, K);
+  } else if (which == 5) {
     if (M % BM || N % BN || K % BK8) return -2;
     dim3 grid(N / BN, M / BM);
     hipLaunchKernelGGL(mfma_gemm_fp8_128u, grid, dim3(256), 0, 0,
