@@ -53,11 +53,11 @@ def test_fp8_deep_kernel_race_screen(attest):
         assert torch.equal(c, ref), f"trial {trial}"
 
 
-@pytest.mark.parametrize("which", [0, 2, 3, 4, 5])
+@pytest.mark.parametrize("which", [0, 2, 3, 4, 5, 6])
 def test_fp8_variants_bitwise(attest, which):
     """Every fp8 structure variant (BK=128 / BK=256 / BK=64-hiocc /
-    BK=128 3-blocks-per-CU / BK=128 single-buffered 4-blocks-per-CU)
-    must agree bitwise with the fp32 reference on integer data."""
+    BK=128 3-blk / BK=128 single-buffered 4-blk / 256x128-tile) must
+    agree bitwise with the fp32 reference on integer data."""
     m = n = 512
     k = 1024
     torch.manual_seed(which)
