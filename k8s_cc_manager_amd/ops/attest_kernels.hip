@@ -1831,12 +1831,13 @@ int cc_attest_device(int device, int gemm_dim, struct CcAttestReport* rep) {
                      elems, 7u);
   {
     dim3 g8(D / BN, D / BM);
-    hipLaunchKernelGGL(mfma_gemm_fp8_128, g8, dim3(256), 0, 0,
+    // the production-dispatch fp8 kernel (4-blocks/CU single-buffered)
+    hipLaunchKernelGGL(mfma_gemm_fp8_128u, g8, dim3(256), 0, 0,
                        (const char*)ctx->dA8, (const char*)ctx->dB8, dC, D, D,
                        D);
     CC_CHECK(hipDeviceSynchronize());
     CC_CHECK(hipEventRecord(ev0, 0));
-    hipLaunchKernelGGL(mfma_gemm_fp8_128, g8, dim3(256), 0, 0,
+    hipLaunchKernelGGL(mfma_gemm_fp8_128u, g8, dim3(256), 0, 0,
                        (const char*)ctx->dA8, (const char*)ctx->dB8, dC, D, D,
                        D);
     CC_CHECK(hipEventRecord(ev1, 0));
