@@ -203,6 +203,7 @@ def main(argv=None) -> int:
                             app="synthetic", gpu_request=1)
 
     def sync() -> None:
+        manager.flush_events()  # async Event posts stay inside the timed region
         if use_gpu:
             torch.cuda.synchronize()
         if world > 1:

@@ -141,7 +141,24 @@ class K8sClient:
 
     def patch_node_labels(self, name: str, labels: Dict[str, Optional[str]]) -> Dict[str, Any]:
         """Merge-patch only the given labels (None deletes a label)."""
-        patch = {"metadata": {"labels": labels}}
+        return self.patch_node(name, labels=labels)
+
+    def patch_node(
+        self,
+        name: str,
+        labels: Optional[Dict[str, Optional[str]]] = None,
+        unschedulable: Optional[bool] = None,
+    ) -> Dict[str, Any]:
+        """One strategic-merge patch combining labels and/or
+        spec.unschedulable — label rewrites and cordon/uncordon land
+        ATOMICALLY in a single API round-trip (no window where the node
+        is paused but schedulable, and one request instead of two on
+        the transition hot path)."""
+        patch: Dict[str, Any] = {}
+        if labels is not None:
+            patch["metadata"] = {"labels": labels}
+        if unschedulable is not None:
+            patch["spec"] = {"unschedulable": unschedulable or None}
         resp = self._request(
             "PATCH",
             f"{self.base_url}/api/v1/nodes/{name}",

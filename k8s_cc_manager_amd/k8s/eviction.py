@@ -108,6 +108,7 @@ def evict_components(
     current_labels: Dict[str, str],
     timeout: float = 300.0,
     poll_interval: float = 2.0,
+    cordon: bool = False,
 ) -> bool:
     """Pause component labels and wait for their pods to drain.
 
@@ -118,11 +119,15 @@ def evict_components(
     """
     paused = {name: pause_value(v) for name, v in current_labels.items()}
     try:
-        k8s.patch_node_labels(node_name, paused)
+        # one atomic patch: pause labels AND cordon together (no window
+        # where components are paused but the node still schedulable)
+        k8s.patch_node(node_name, labels=paused,
+                       unschedulable=True if cordon else None)
     except ApiError as e:
         logger.error("failed to pause component labels: %s", e)
         return False
-    logger.info("paused %d component labels", len(paused))
+    logger.info("paused %d component labels%s", len(paused),
+                " + cordoned" if cordon else "")
 
     # Components that actually had pods to drain: deployed (non-empty,
     # non-'false') values only.
@@ -138,22 +143,29 @@ def evict_components(
     deadline = time.monotonic() + timeout
     delay = min(0.002, poll_interval)
     while pending and time.monotonic() < deadline:
-        for app in sorted(pending):
-            try:
-                pods = k8s.list_pods(
-                    operator_namespace,
-                    field_selector=f"spec.nodeName={node_name}",
-                    label_selector=f"app={app}",
-                )
-                n = len(pods.get("items") or [])
-            except ApiError as e:
-                logger.warning("pod-drain poll error for %s: %s", app, e)
-                continue
-            if n == 0:
-                logger.info("%s drained", app)
-                pending.discard(app)
-            else:
-                logger.debug("%s: %d pod(s) remaining", app, n)
+        # ONE list per poll round (node-scoped), filtered client-side —
+        # the per-component poll the reference does (g_o_e.py:189-204)
+        # costs N API calls per round; pod counts per node are small,
+        # so the filtering is cheaper than the round-trips.
+        try:
+            pods = k8s.list_pods(
+                operator_namespace,
+                field_selector=f"spec.nodeName={node_name}",
+            )
+            present = {
+                ((p.get("metadata") or {}).get("labels") or {}).get("app")
+                for p in pods.get("items") or []
+            }
+        except ApiError as e:
+            logger.warning("pod-drain poll error: %s", e)
+            time.sleep(delay)
+            delay = min(delay * 2, poll_interval)
+            continue
+        for app in sorted(pending & present):
+            logger.debug("%s: pod(s) remaining", app)
+        for app in sorted(pending - present):
+            logger.info("%s drained", app)
+        pending &= present
         if pending:
             time.sleep(delay)
             delay = min(delay * 2, poll_interval)
@@ -164,26 +176,36 @@ def evict_components(
 
 
 def reschedule_components(
-    k8s: K8sClient, node_name: str, original_labels: Dict[str, str]
+    k8s: K8sClient, node_name: str, original_labels: Dict[str, str],
+    uncordon: bool = False,
+    extra_labels: Optional[Dict[str, str]] = None,
 ) -> bool:
-    """Restore component labels so the operator reschedules the pods."""
+    """Restore component labels so the operator reschedules the pods
+    (optionally uncordoning and publishing ``extra_labels`` — e.g. the
+    post-transition state pair — in the same atomic patch)."""
     restored = {name: unpause_value(v) for name, v in original_labels.items()}
+    if extra_labels:
+        restored.update(extra_labels)
     try:
-        k8s.patch_node_labels(node_name, restored)
+        k8s.patch_node(node_name, labels=restored,
+                       unschedulable=False if uncordon else None)
     except ApiError as e:
         logger.error("failed to restore component labels: %s", e)
         return False
-    logger.info("restored %d component labels", len(restored))
+    logger.info("restored %d component labels%s", len(restored),
+                " + uncordoned" if uncordon else "")
     return True
 
 
 def unwind_paused_labels(
-    k8s: K8sClient, node_name: str, original_labels: Dict[str, str]
+    k8s: K8sClient, node_name: str, original_labels: Dict[str, str],
+    uncordon: bool = False,
 ) -> None:
     """Best-effort restore after a failed eviction (reference gap:
     main.py:558-566 leaves components paused on that path)."""
     try:
-        reschedule_components(k8s, node_name, original_labels)
+        reschedule_components(k8s, node_name, original_labels,
+                              uncordon=uncordon)
     except Exception as e:  # pragma: no cover - double fault
         logger.error("could not unwind paused labels: %s", e)
 
@@ -261,14 +283,17 @@ def evict_gpu_workload_pods(
     return ok
 
 
+def state_label_dict(state: str) -> Dict[str, str]:
+    """mode.state + derived ready.state label pair for ``state``."""
+    return {CC_STATE_LABEL: state, CC_READY_LABEL: ready_value_for_state(state)}
+
+
 def set_cc_state_label(k8s: K8sClient, node_name: str, state: str) -> bool:
     """Publish mode.state + derived ready.state (reference semantics,
     gpu_operator_eviction.py:262-295)."""
     ready = ready_value_for_state(state)
     try:
-        k8s.patch_node_labels(
-            node_name, {CC_STATE_LABEL: state, CC_READY_LABEL: ready}
-        )
+        k8s.patch_node_labels(node_name, state_label_dict(state))
     except ApiError as e:
         logger.error("failed to set state labels: %s", e)
         return False

@@ -14,18 +14,29 @@ NODE = "node0"
 
 
 class PatchFaultClient:
-    """Fails patch_node_labels calls that try to PAUSE components."""
+    """Fails node patches that try to PAUSE components (the combined
+    cordon+pause patch is atomic, so failing it leaves NOTHING
+    applied — the property the test asserts)."""
 
     def __init__(self, inner: K8sClient):
         self.inner = inner
         self.fail_pause = True
 
-    def patch_node_labels(self, name, labels):
-        if self.fail_pause and any(
+    def _is_pause(self, labels):
+        return labels and any(
             isinstance(v, str) and PAUSED_VALUE in v for v in labels.values()
-        ):
+        )
+
+    def patch_node_labels(self, name, labels):
+        if self.fail_pause and self._is_pause(labels):
             raise ApiError(500, "injected pause failure")
         return self.inner.patch_node_labels(name, labels)
+
+    def patch_node(self, name, labels=None, unschedulable=None):
+        if self.fail_pause and self._is_pause(labels):
+            raise ApiError(500, "injected pause failure")
+        return self.inner.patch_node(name, labels=labels,
+                                     unschedulable=unschedulable)
 
     def __getattr__(self, item):
         return getattr(self.inner, item)

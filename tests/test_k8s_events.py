@@ -25,6 +25,7 @@ def test_success_events(fake_cluster):
     cluster.add_node(NODE)
     mgr = _mgr(cluster, url, MockBackend(num_gpus=2))
     assert mgr.apply_mode("on")
+    mgr.flush_events()
     reasons = [e["reason"] for e in cluster.k8s_events]
     assert reasons == ["CCTransitionStarted", "CCTransitionSucceeded"]
     done = cluster.k8s_events[-1]
@@ -41,6 +42,7 @@ def test_failure_event_is_warning(fake_cluster):
     be = MockBackend(num_gpus=2, faults=FaultPlan(fail_reset=["0000:10:00.0"]))
     mgr = _mgr(cluster, url, be)
     assert not mgr.apply_mode("on")
+    mgr.flush_events()
     last = cluster.k8s_events[-1]
     assert last["reason"] == "CCTransitionFailed"
     assert last["type"] == "Warning"
@@ -52,4 +54,5 @@ def test_idempotent_apply_emits_no_transition_events(fake_cluster):
     cluster.add_node(NODE)
     mgr = _mgr(cluster, url, MockBackend(num_gpus=1, initial_cc_mode="on"))
     assert mgr.apply_mode("on")
+    mgr.flush_events()
     assert cluster.k8s_events == []
