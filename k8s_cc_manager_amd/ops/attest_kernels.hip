@@ -1394,10 +1394,20 @@ __global__ void checksum_f32(const float* __restrict__ x, long n,
     v ^= v >> 33; v *= 0xff51afd7ed558ccdull; v ^= v >> 33;
     h ^= v;
   }
-  // xor-reduce across the wave then one atomic per wave
+  // xor-reduce wave -> block (LDS) -> ONE atomic per block. The first
+  // version did one atomic per wave: 4096 serialized atomicXors on a
+  // single address cost ~40 us — more than the whole hash sweep
+  // (bench kernel trace, profiles/bench_kernel_stats.csv).
   for (int off = 32; off > 0; off >>= 1)
     h ^= __shfl_down(h, off, 64);
-  if ((threadIdx.x & 63) == 0) atomicXor(out, h);
+  __shared__ unsigned long long partial[4];
+  if ((threadIdx.x & 63) == 0) partial[threadIdx.x >> 6] = h;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    unsigned long long b = partial[0];
+    for (int w = 1; w < (int)(blockDim.x >> 6); ++w) b ^= partial[w];
+    atomicXor(out, b);
+  }
 }
 
 // ---------------------------------------------------------------------------
