@@ -49,6 +49,30 @@ def test_failure_event_is_warning(fake_cluster):
     assert "0000:10:00.0" in last["message"]
 
 
+def test_event_worker_survives_post_failures(fake_cluster):
+    """A failing Events endpoint must not wedge the worker or the
+    queue — posts are best-effort observability."""
+    cluster, url = fake_cluster
+    cluster.add_node(NODE)
+    mgr = _mgr(cluster, url, MockBackend(num_gpus=1))
+
+    calls = {"n": 0}
+
+    def failing_create_event(*a, **kw):
+        calls["n"] += 1
+        raise RuntimeError("injected event failure")
+
+    mgr.k8s.create_event = failing_create_event
+    assert mgr.apply_mode("on")  # transition unaffected
+    mgr.flush_events(timeout=5.0)
+    assert calls["n"] == 2  # both events were attempted
+    assert mgr._event_q.unfinished_tasks == 0  # queue fully drained
+    # worker is still alive for subsequent events
+    assert mgr.apply_mode("off")
+    mgr.flush_events(timeout=5.0)
+    assert calls["n"] == 4
+
+
 def test_idempotent_apply_emits_no_transition_events(fake_cluster):
     cluster, url = fake_cluster
     cluster.add_node(NODE)
