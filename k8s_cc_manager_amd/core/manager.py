@@ -65,6 +65,14 @@ class ManagerConfig:
         import os
 
         env = env if env is not None else os.environ
+
+        def _f(name: str, default: float) -> float:
+            try:
+                return float(env.get(name, default))
+            except ValueError:
+                logger.warning("ignoring non-numeric %s=%r", name, env.get(name))
+                return default
+
         return cls(
             operator_namespace=env.get("OPERATOR_NAMESPACE", "amd-gpu-operator"),
             evict_components=env.get("EVICT_OPERATOR_COMPONENTS", "true").lower()
@@ -73,6 +81,13 @@ class ManagerConfig:
             == "true",
             cordon_node=env.get("CORDON_NODE", "true").lower() == "true",
             readiness_file=env.get("CC_READINESS_FILE"),
+            # timing envelope: the reference hardcodes these
+            # (g_o_e.py:136,200; main.py:628,684) — here they are
+            # tunable per deployment
+            eviction_timeout=_f("CC_EVICTION_TIMEOUT", 300.0),
+            eviction_poll_interval=_f("CC_EVICTION_POLL_INTERVAL", 2.0),
+            watch_timeout_seconds=int(_f("CC_WATCH_TIMEOUT", 300)),
+            reconnect_backoff=_f("CC_RECONNECT_BACKOFF", 5.0),
         )
 
 

@@ -43,6 +43,18 @@ def test_manager_config_from_env(monkeypatch):
     assert cfg.cordon_node is False
 
 
+def test_manager_config_timing_env(monkeypatch):
+    monkeypatch.setenv("CC_EVICTION_TIMEOUT", "45.5")
+    monkeypatch.setenv("CC_EVICTION_POLL_INTERVAL", "0.25")
+    monkeypatch.setenv("CC_WATCH_TIMEOUT", "60")
+    monkeypatch.setenv("CC_RECONNECT_BACKOFF", "bogus")  # ignored
+    cfg = ManagerConfig.from_env()
+    assert cfg.eviction_timeout == 45.5
+    assert cfg.eviction_poll_interval == 0.25
+    assert cfg.watch_timeout_seconds == 60
+    assert cfg.reconnect_backoff == 5.0  # default kept on bad value
+
+
 def test_kubeconfig_loading(fake_cluster, tmp_path):
     cluster, url = fake_cluster
     cluster.add_node("kcnode", labels={"x": "1"})
