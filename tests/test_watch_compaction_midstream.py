@@ -27,19 +27,25 @@ def test_midstream_compaction_emits_410():
     def watcher():
         # deliberately SLOW consumer: backpressure lets the server-side
         # event log trim past this watcher's cursor
-        for event in k8s.watch_node(NODE, resource_version=rv, timeout_seconds=15):
+        for event in k8s.watch_node(NODE, resource_version=rv, timeout_seconds=30):
             seen.append(event)
             if event.get("type") == "ERROR":
                 return
-            time.sleep(0.05)
+            time.sleep(0.005)
 
     t = threading.Thread(target=watcher)
     t.start()
     time.sleep(0.2)
-    # burst enough churn to trim the event log past the watcher's cursor
-    for i in range(1500):
-        cluster.set_node_label(NODE, "churn", str(i))
-    t.join(timeout=30)
+    # Burst enough churn BYTES that the server-side handler must block
+    # on the socket (the consumer is slower than the producer): kernel
+    # socket buffers can absorb hundreds of KB, so the burst has to be
+    # megabytes — a small burst fits in the buffers, the handler never
+    # falls behind the trim, and no 410 is provoked (seen as a flake on
+    # fast boxes).
+    pad = "x" * 120
+    for i in range(30000):
+        cluster.set_node_label(NODE, "churn", f"{pad}{i}")
+    t.join(timeout=60)
     cluster.stop()
     assert any(
         e.get("type") == "ERROR" and e["object"].get("code") == 410 for e in seen
