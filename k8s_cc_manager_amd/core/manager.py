@@ -21,7 +21,7 @@ import time
 from dataclasses import dataclass
 from typing import Optional
 
-from ..device.contract import DeviceBackend
+from ..device.contract import FABRIC_OFF, DeviceBackend
 from ..k8s import eviction
 from ..k8s.client import ApiError, K8sClient
 from ..labels import (
@@ -244,7 +244,14 @@ class CCManager:
             self._set_state(MODE_OFF)
             return True
 
-        if self._cc_mode_is_set(cc_gpus, mode):
+        # Idempotency pre-check. Unlike the reference's mode_is_set
+        # (main.py:427-446), this also requires the FABRIC-protected
+        # mode to be off: under ppcie the CC register reads 'off', so a
+        # ppcie->off label flip would otherwise short-circuit here,
+        # publish state=off, and leave the xGMI hive protected — a
+        # half-transitioned node the reference mislabels (its phase 1
+        # at main.py:473-483 never runs when the pre-check passes).
+        if self._cc_mode_is_set(cc_gpus, mode) and self._fabric_all_off():
             logger.info("all GPUs already in CC mode %r", mode)
             self._set_state(mode)
             return True
@@ -278,6 +285,20 @@ class CCManager:
             return all(g.query_cc_mode() == mode for g in gpus)
         except Exception as e:
             logger.error("CC mode pre-check failed: %s", e)
+            return False
+
+    def _fabric_all_off(self) -> bool:
+        """No device may still be in fabric-protected mode for a CC-mode
+        idempotency short-circuit to be valid (see _apply_mode_locked)."""
+        try:
+            devices, _ = self.backend.find_devices()
+            return all(
+                d.query_fabric_mode() == FABRIC_OFF
+                for d in devices
+                if d.fabric_query_supported
+            )
+        except Exception as e:
+            logger.error("fabric-off pre-check failed: %s", e)
             return False
 
     def _fabric_mode_is_set(self, devices) -> bool:

@@ -227,3 +227,34 @@ def test_watch_resyncs_after_compaction(node_cluster):
     assert cluster.node_labels(NODE).get(CC_STATE_LABEL) == "devtools"
     mgr.stop_event.set()
     t.join(timeout=5)
+
+
+def test_ppcie_to_off_disables_fabric(fake_cluster):
+    """ppcie -> off must actually run the transition and force fabric
+    mode off. The reference's idempotency pre-check (main.py:427-446)
+    only reads the CC register — which is 'off' under ppcie — so it
+    short-circuits and leaves the hive protected while publishing
+    state=off. This pre-check also requires fabric-off."""
+    from k8s_cc_manager_amd.labels import CC_STATE_LABEL
+
+    cluster, url = fake_cluster
+    cluster.add_node(NODE)
+    backend = MockBackend(num_gpus=2)
+    mgr = CCManager(
+        node_name=NODE,
+        default_mode="off",
+        host_cc=True,
+        k8s=K8sClient(url),
+        backend=backend,
+        engine=TransitionEngine(),
+        config=ManagerConfig(evict_components=False, cordon_node=False),
+    )
+    assert mgr.apply_mode("ppcie")
+    assert all(d.query_fabric_mode() == "on" for d in backend.get_gpus())
+    assert mgr.apply_mode("off")
+    mgr.flush_events()
+    assert all(d.query_fabric_mode() == "off" for d in backend.get_gpus()), (
+        "fabric left protected after ppcie->off"
+    )
+    assert all(d.query_cc_mode() == "off" for d in backend.get_gpus())
+    assert cluster.node_labels(NODE)[CC_STATE_LABEL] == "off"
