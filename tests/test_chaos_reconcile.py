@@ -26,7 +26,7 @@ NODE = "chaos0"
 MODES = ["on", "off", "devtools", "ppcie"]
 
 
-@pytest.mark.parametrize("seed", [7, 1234, 987654])
+@pytest.mark.parametrize("seed", [7, 1234, 987654, 5150])
 def test_chaos_converges(seed):
     rng = random.Random(seed)
     cluster = FakeCluster(event_log_max=300, operator_tick=0.01)
