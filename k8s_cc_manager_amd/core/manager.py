@@ -466,11 +466,30 @@ class CCManager:
                         self._node_labels_cache = dict(labels)
                         self.current_label = labels.get(CC_MODE_LABEL, "")
                         if self.current_label != last_applied:
-                            logger.info(
-                                "label changed %r -> %r", last_applied, self.current_label
-                            )
-                            last_applied = self.current_label
-                            self.apply_mode(self.with_default(self.current_label))
+                            # COALESCE label flaps: a burst of flips
+                            # buffered behind a long blocking apply
+                            # would otherwise replay every intermediate
+                            # transition (each a full evict+reset cycle
+                            # — minutes on real hardware; the reference
+                            # replays them all, main.py:646-657).
+                            # Confirm against an authoritative read and
+                            # apply only the LATEST value; stale
+                            # buffered events then match last_applied
+                            # and are skipped, or collapse into the
+                            # idempotency pre-check.
+                            fresh = self.current_label
+                            try:
+                                fresh = self.read_mode_label()
+                            except ApiError as e:
+                                logger.warning(
+                                    "confirm-read failed (%s); using event label", e
+                                )
+                            if fresh != last_applied:
+                                logger.info(
+                                    "label changed %r -> %r", last_applied, fresh
+                                )
+                                last_applied = fresh
+                                self.apply_mode(self.with_default(fresh))
                 if resync:
                     last_applied = self._resync(last_applied)
             except ApiError as e:
