@@ -110,12 +110,14 @@ def evict_components(
     poll_interval: float = 2.0,
     cordon: bool = False,
 ) -> bool:
-    """Pause component labels and wait for their pods to drain.
+    """Pause component labels (atomically with the cordon when
+    ``cordon``) and wait for their pods to drain.
 
     Returns True when every deployed component's pods are gone (or the
     drain deadline passed — drain timeout is non-fatal, matching the
-    reference envelope g_o_e.py:205-207). Returns False after UNWINDING
-    the labels if the API rejects the pause patch.
+    reference envelope g_o_e.py:205-207). Returns False if the API
+    rejects the pause patch — the patch is atomic, so a False return
+    means NOTHING was applied (no unwind needed).
     """
     paused = {name: pause_value(v) for name, v in current_labels.items()}
     try:
