@@ -227,6 +227,8 @@ class CCManager:
 
         gpus = self.backend.get_gpus()
         cc_gpus = self.backend.get_cc_capable_gpus()
+        if METRICS.enabled:
+            METRICS.devices_managed.set(len(cc_gpus))
 
         # Mixed capability with a non-off target is unrecoverable
         # (reference main.py:237-240).

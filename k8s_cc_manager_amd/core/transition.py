@@ -34,6 +34,7 @@ from ..device.contract import (
     ModeVerifyError,
 )
 from ..parallel.executor import DeviceExecutor
+from ..utils.metrics import METRICS
 from ..utils.timing import PhaseTimer
 
 logger = logging.getLogger(__name__)
@@ -80,7 +81,12 @@ class TransitionEngine:
                 f"{dev.bdf}: {what} mode readback {got!r} != staged {want!r}"
             )
         if self.attestor is not None:
-            self.attestor(dev)
+            try:
+                self.attestor(dev)
+            except Exception:
+                if METRICS.enabled:
+                    METRICS.attest_failures.inc()
+                raise
 
     # ------------------------------------------------------------------
     def apply_cc_mode(
