@@ -256,12 +256,20 @@ def evict_gpu_workload_pods(
     logger.info("evicting %d GPU workload pod(s): %s", len(targets),
                 [f"{ns}/{n}" for ns, n in targets])
     ok = True
+    retry_429: set = set()
     for ns, name in targets:
         try:
             k8s.evict_pod(ns, name)
         except ApiError as e:
-            logger.warning("eviction of %s/%s rejected: %s", ns, name, e)
-            ok = False
+            if e.status == 429:
+                # PodDisruptionBudget temporarily blocks this eviction
+                # (k8s contract: retry later) — keep retrying inside
+                # the drain deadline instead of failing outright
+                logger.info("eviction of %s/%s blocked by PDB; will retry", ns, name)
+                retry_429.add((ns, name))
+            else:
+                logger.warning("eviction of %s/%s rejected: %s", ns, name, e)
+                ok = False
 
     deadline = time.monotonic() + timeout
     delay = min(0.002, poll_interval)
@@ -276,9 +284,22 @@ def evict_gpu_workload_pods(
             remaining &= alive
         except ApiError as e:
             logger.warning("GPU-workload drain poll error: %s", e)
+        for ns, name in sorted(retry_429 & remaining):
+            try:
+                k8s.evict_pod(ns, name)
+                retry_429.discard((ns, name))
+            except ApiError as e:
+                if e.status != 429:
+                    logger.warning("eviction retry of %s/%s rejected: %s", ns, name, e)
+                    retry_429.discard((ns, name))
+                    ok = False
         if remaining:
             time.sleep(delay)
             delay = min(delay * 2, poll_interval)
+    if retry_429 & remaining:
+        logger.warning("PDB still blocking eviction at deadline: %s",
+                       sorted(retry_429 & remaining))
+        ok = False
     if remaining:
         logger.warning("GPU workload pods still terminating at deadline: %s",
                        sorted(remaining))

@@ -70,6 +70,8 @@ class FakeCluster:
         self._pending_delete: Dict[Tuple[str, str], float] = {}
         self._pending_evict: Dict[Tuple[str, str, str], float] = {}
         self._evictions: List[Tuple[str, str]] = []
+        #: (ns, name) -> remaining 429 responses (PDB-block simulation)
+        self._evict_429: Dict[Tuple[str, str], int] = {}
         self.k8s_events: List[Dict[str, Any]] = []
         self._events: List[Dict[str, Any]] = []  # {"rv", "type", "node"}
 
@@ -152,6 +154,12 @@ class FakeCluster:
                     continue
                 out.append(json.loads(json.dumps(pod)))
             return out
+
+    def block_eviction(self, namespace: str, name: str, times: int) -> None:
+        """Make the next ``times`` Eviction POSTs for this pod answer
+        429 (PodDisruptionBudget simulation)."""
+        with self._lock:
+            self._evict_429[(namespace, name)] = times
 
     def compact(self) -> None:
         """Mark all current events compacted -> old-RV watches get 410."""
@@ -325,6 +333,17 @@ class FakeCluster:
                         )
                         if key is None:
                             return self._send_json(404, {"kind": "Status", "code": 404})
+                        left = cluster._evict_429.get((ns, name), 0)
+                        if left > 0:
+                            # PodDisruptionBudget temporarily blocks this
+                            # eviction (the real API answers 429)
+                            cluster._evict_429[(ns, name)] = left - 1
+                            return self._send_json(
+                                429,
+                                {"kind": "Status", "code": 429,
+                                 "reason": "TooManyRequests",
+                                 "message": "disruption budget blocked"},
+                            )
                         cluster._evictions.append((ns, name))
                         if cluster.delete_delay <= 0:
                             cluster._pods.pop(key, None)

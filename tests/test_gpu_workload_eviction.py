@@ -97,3 +97,31 @@ def test_config_from_env(monkeypatch):
     assert ManagerConfig.from_env().evict_gpu_workloads is True
     monkeypatch.delenv("EVICT_GPU_WORKLOADS")
     assert ManagerConfig.from_env().evict_gpu_workloads is False
+
+
+def test_pdb_429_is_retried_until_allowed(fake_cluster):
+    """The Eviction API answers 429 while a PodDisruptionBudget blocks
+    the eviction (k8s contract: retry later). The drain loop must keep
+    retrying inside the deadline and succeed once the budget allows."""
+    cluster, url = fake_cluster
+    cluster.add_node(NODE)
+    cluster.add_pod("user-ns", "pdb-pod", NODE, app="trainer", gpu_request=1)
+    cluster.block_eviction("user-ns", "pdb-pod", times=3)
+    k8s = K8sClient(url)
+    assert evict_gpu_workload_pods(k8s, NODE, timeout=5.0, poll_interval=0.02)
+    assert ("user-ns", "pdb-pod") in cluster._evictions
+    assert cluster.pods_on(NODE) == []
+
+
+def test_pdb_429_forever_fails_at_deadline(fake_cluster):
+    """A budget that never allows the eviction: bounded failure at the
+    drain deadline (returns False, pod still present, loudly logged)."""
+    cluster, url = fake_cluster
+    cluster.add_node(NODE)
+    cluster.add_pod("user-ns", "stuck-pdb", NODE, app="trainer", gpu_request=1)
+    cluster.block_eviction("user-ns", "stuck-pdb", times=10**6)
+    k8s = K8sClient(url)
+    t0 = time.monotonic()
+    assert not evict_gpu_workload_pods(k8s, NODE, timeout=0.5, poll_interval=0.02)
+    assert time.monotonic() - t0 < 5
+    assert any(p["metadata"]["name"] == "stuck-pdb" for p in cluster.pods_on(NODE))
