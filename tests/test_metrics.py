@@ -41,3 +41,26 @@ def test_transition_metrics_recorded(fake_cluster):
     # histogram recorded at least one observation for the whole transition
     count = _sample("cc_transition_seconds_count", {"mode": "on"})
     assert count and count >= 1
+
+
+def test_attest_failure_metric_increments():
+    """A failing attestor must bump cc_attest_failures_total."""
+    import pytest
+
+    from k8s_cc_manager_amd.core.transition import TransitionEngine
+    from k8s_cc_manager_amd.device.mock import MockBackend
+    from k8s_cc_manager_amd.utils.metrics import METRICS
+
+    if not METRICS.enabled:
+        pytest.skip("prometheus_client unavailable")
+
+    def bad_attestor(dev):
+        raise RuntimeError("injected attestation failure")
+
+    backend = MockBackend(num_gpus=1)
+    engine = TransitionEngine(attestor=bad_attestor)
+    before = METRICS.attest_failures._value.get()
+    report = engine.apply_cc_mode(backend.get_gpus(), backend.get_gpus(), "on")
+    assert not report.ok
+    assert "attestation" in report.error
+    assert METRICS.attest_failures._value.get() == before + 1
