@@ -1639,8 +1639,11 @@ int cc_mfma_gemm_bf16(int device, const void* A, const void* Bt, void* C,
   return (int)hipDeviceSynchronize();
 }
 
-// Force a specific variant (perf characterization / A-B): which=0 the
-// 128x128 step-3 kernel, which=1 the 256x256 8-phase kernel.
+// Force a specific bf16 variant (perf characterization / A-B):
+// 0 = 128x128 step-3, 1 = 256x256 8-phase (16x16 op),
+// 2 = 256x256 8-phase (32x32 op), 3 = 128x128 single-buffered
+// 4-blocks/CU (measured slower than the deep pipeline — kept as the
+// recorded negative arm of the occupancy series).
 int cc_mfma_gemm_bf16_variant(int device, const void* A, const void* Bt,
                               void* C, int M, int N, int K, int which) {
   if (hipSetDevice(device) != hipSuccess) return -3;
@@ -1671,8 +1674,13 @@ int cc_mfma_gemm_bf16_variant(int device, const void* A, const void* Bt,
   return (int)hipDeviceSynchronize();
 }
 
-// Force a specific fp8 variant: 0 = 128-tile BK=128 step-3,
-// 1 = 256-tile deep pipeline, 2 = 128-tile BK=256 step-3.
+// Force a specific fp8 variant (perf characterization / A-B):
+// 0 = 128-tile BK=128 double-buffered (2 blocks/CU),
+// 1 = 256-tile deep pipeline, 2 = 128-tile BK=256 (1 block/CU),
+// 3 = BK=64 32x32-op (4 blocks/CU), 4 = BK=128 1.5-buffered
+// (3 blocks/CU), 5 = BK=128 SINGLE-buffered (4 blocks/CU — the
+// production dispatch winner), 6 = 256x128 tile, 512 threads
+// (2 blocks/CU). Measured ladder in BASELINE.md.
 int cc_mfma_gemm_fp8_variant(int device, const void* A, const void* Bt,
                              void* C, int M, int N, int K, int which) {
   if (hipSetDevice(device) != hipSuccess) return -3;
