@@ -137,3 +137,38 @@ def test_watch_error_budget_exhausts(fake_cluster):
     with pytest.raises(RuntimeError, match="3 times consecutively"):
         mgr.run()
     assert failing.watch_calls == 3
+
+
+def test_kubeconfig_client_cert_data_materialized(tmp_path):
+    """client-certificate-data / client-key-data (base64 inline) must be
+    materialized to temp PEM files and wired as the client cert pair."""
+    import base64
+    import os
+
+    import yaml
+
+    from k8s_cc_manager_amd.k8s.client import K8sClient
+
+    cert_pem = b"-----BEGIN CERTIFICATE-----\nZZZZ\n-----END CERTIFICATE-----\n"
+    key_pem = b"-----BEGIN PRIVATE KEY-----\nYYYY\n-----END PRIVATE KEY-----\n"
+    kc = {
+        "apiVersion": "v1",
+        "current-context": "c",
+        "contexts": [{"name": "c", "context": {"cluster": "cl", "user": "u"}}],
+        "clusters": [{"name": "cl", "cluster": {
+            "server": "https://example.invalid:6443",
+            "insecure-skip-tls-verify": True,
+        }}],
+        "users": [{"name": "u", "user": {
+            "client-certificate-data": base64.b64encode(cert_pem).decode(),
+            "client-key-data": base64.b64encode(key_pem).decode(),
+        }}],
+    }
+    path = tmp_path / "kubeconfig"
+    path.write_text(yaml.safe_dump(kc))
+    client = K8sClient.from_kubeconfig(str(path))
+    assert client._verify is False
+    assert client._cert is not None
+    cc, ck = client._cert
+    assert os.path.exists(cc) and open(cc, "rb").read() == cert_pem
+    assert os.path.exists(ck) and open(ck, "rb").read() == key_pem
