@@ -1,0 +1,48 @@
+"""Direct unit tests of the fake API server's strategic-merge-patch
+semantics (everything else rides on these: label deletes via null,
+nested merges, spec.unschedulable lifecycle)."""
+
+from k8s_cc_manager_amd.k8s.client import K8sClient
+from k8s_cc_manager_amd.k8s.fakecluster import _merge_patch
+
+
+def test_merge_patch_null_deletes():
+    target = {"metadata": {"labels": {"a": "1", "b": "2"}}}
+    _merge_patch(target, {"metadata": {"labels": {"a": None, "c": "3"}}})
+    assert target["metadata"]["labels"] == {"b": "2", "c": "3"}
+
+
+def test_merge_patch_replaces_non_dict_with_dict():
+    target = {"spec": "bogus"}
+    _merge_patch(target, {"spec": {"unschedulable": True}})
+    assert target["spec"] == {"unschedulable": True}
+
+
+def test_merge_patch_scalar_overwrite_and_deep_merge():
+    target = {"a": {"b": {"c": 1, "keep": "x"}}, "top": 1}
+    _merge_patch(target, {"a": {"b": {"c": 2}}, "top": 2})
+    assert target == {"a": {"b": {"c": 2, "keep": "x"}}, "top": 2}
+
+
+def test_label_delete_roundtrip_over_http(fake_cluster):
+    """Deleting a label with a null value through the real HTTP path."""
+    cluster, url = fake_cluster
+    cluster.add_node("n0", labels={"keep": "1", "drop": "2"})
+    k8s = K8sClient(url)
+    k8s.patch_node_labels("n0", {"drop": None, "new": "3"})
+    labels = cluster.node_labels("n0")
+    assert labels == {"keep": "1", "new": "3"}
+
+
+def test_uncordon_removes_unschedulable_key(fake_cluster):
+    """Cordon sets spec.unschedulable; uncordon must clear it the way
+    kubectl does (the field disappears rather than reading false)."""
+    cluster, url = fake_cluster
+    cluster.add_node("n0")
+    k8s = K8sClient(url)
+    k8s.patch_node("n0", unschedulable=True)
+    assert cluster.node_unschedulable("n0")
+    k8s.patch_node("n0", unschedulable=False)
+    assert not cluster.node_unschedulable("n0")
+    node = cluster.get_node_copy("n0")
+    assert "unschedulable" not in node["spec"]
