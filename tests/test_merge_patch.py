@@ -46,3 +46,31 @@ def test_uncordon_removes_unschedulable_key(fake_cluster):
     assert not cluster.node_unschedulable("n0")
     node = cluster.get_node_copy("n0")
     assert "unschedulable" not in node["spec"]
+
+
+def test_merge_patch_idempotent_property():
+    """Applying the same patch twice is a no-op the second time."""
+    import copy
+
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    scalars = st.one_of(st.none(), st.integers(-5, 5), st.text(max_size=4))
+    patches = st.recursive(
+        st.dictionaries(st.text(max_size=3), scalars, max_size=3),
+        lambda children: st.dictionaries(
+            st.text(max_size=3), st.one_of(scalars, children), max_size=3
+        ),
+        max_leaves=8,
+    )
+
+    @settings(max_examples=200, deadline=None)
+    @given(patches, patches)
+    def check(base, patch):
+        target = copy.deepcopy(base)
+        _merge_patch(target, patch)
+        once = copy.deepcopy(target)
+        _merge_patch(target, patch)
+        assert target == once
+
+    check()
