@@ -87,3 +87,42 @@ def test_https_watch_stream(tls_cluster):
         if event["object"]["metadata"]["labels"].get("watched") == "yes":
             break
     assert seen
+
+
+def test_full_reconcile_over_tls(tls_cluster):
+    """End-to-end reconcile (eviction + atomic patches + events) over
+    the in-cluster-shaped path: HTTPS with CA verify + bearer token."""
+    from k8s_cc_manager_amd.core.manager import CCManager, ManagerConfig
+    from k8s_cc_manager_amd.core.transition import TransitionEngine
+    from k8s_cc_manager_amd.device.mock import MockBackend
+    from k8s_cc_manager_amd.k8s.eviction import COMPONENT_LABELS
+    from k8s_cc_manager_amd.labels import CC_READY_LABEL, CC_STATE_LABEL
+
+    cluster, url, ca = tls_cluster
+    for name in COMPONENT_LABELS:
+        cluster.set_node_label("tlsnode", name, "true")
+    mgr = CCManager(
+        node_name="tlsnode",
+        default_mode="on",
+        host_cc=True,
+        k8s=K8sClient(url, token="sekrit", verify=ca),
+        backend=MockBackend(num_gpus=2),
+        engine=TransitionEngine(),
+        config=ManagerConfig(
+            evict_components=True,
+            cordon_node=True,
+            eviction_timeout=5.0,
+            eviction_poll_interval=0.02,
+        ),
+    )
+    mgr.read_mode_label()
+    assert mgr.apply_mode("on")
+    mgr.flush_events()
+    labels = cluster.node_labels("tlsnode")
+    assert labels[CC_STATE_LABEL] == "on"
+    assert labels[CC_READY_LABEL] == "true"
+    assert all(labels[n] == "true" for n in COMPONENT_LABELS)
+    assert not cluster.node_unschedulable("tlsnode")
+    assert [e["reason"] for e in cluster.k8s_events][-2:] == [
+        "CCTransitionStarted", "CCTransitionSucceeded",
+    ]
