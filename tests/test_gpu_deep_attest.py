@@ -2,7 +2,12 @@
 
 import pytest
 
-pytestmark = [pytest.mark.gpu]
+torch = pytest.importorskip("torch")
+
+pytestmark = [
+    pytest.mark.gpu,
+    pytest.mark.skipif(not torch.cuda.is_available(), reason="needs an MI355X"),
+]
 
 
 def test_deep_attest_counts_mfma_cycles():

@@ -2,8 +2,12 @@
 
 import pytest
 
-gpu = pytest.mark.gpu
-pytestmark = [gpu]
+torch = pytest.importorskip("torch")
+
+pytestmark = [
+    pytest.mark.gpu,
+    pytest.mark.skipif(not torch.cuda.is_available(), reason="needs an MI355X"),
+]
 
 
 def test_native_pci_scan_finds_amd_gpu():

@@ -7,8 +7,13 @@ from pathlib import Path
 
 import pytest
 
+torch = pytest.importorskip("torch")
+
 REPO = Path(__file__).resolve().parent.parent
-pytestmark = [pytest.mark.gpu]
+pytestmark = [
+    pytest.mark.gpu,
+    pytest.mark.skipif(not torch.cuda.is_available(), reason="needs an MI355X"),
+]
 
 
 def test_doctor_attest_on_gpu():
