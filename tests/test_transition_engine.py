@@ -136,9 +136,11 @@ def test_fabric_force_off_first():
 
 def test_concurrent_transition_64_devices_randomized_latency():
     """Phase-gather correctness well past node scale: 64 mock GPUs with
-    randomized per-device latencies; the stage-all/reset-all invariant
-    must hold (no device may observe its own reset before every device
-    staged) and all 64 must land on the target mode."""
+    randomized per-device latencies must all land on the target mode,
+    with each device's own op log showing stage strictly before reset
+    (the cross-device stage-ALL-before-reset-ANY invariant is enforced
+    structurally by the executor's phase gather, asserted separately in
+    test_stage_all_before_reset_all)."""
     import random
 
     from k8s_cc_manager_amd.device.mock import MockBackend, MockLatency
